@@ -1,0 +1,66 @@
+"""ResNet experiment family.
+
+Capability parity with the reference's slim cross-product registry
+(experiments/slims.py:164-196, ``slim-<model>-<dataset>``): every ResNet in
+the model zoo is registered against the ``imagenet`` (3x224x224, 1000
+classes) and ``cifar10`` (3x32x32, 10 classes) dataset shapes, under both a
+concise name (``resnet50-imagenet``) and the reference's slim alias
+(``slim-resnet_v1_50-imagenet``). Data is synthetic (BASELINE.json's big
+configs are synthetic/random-init by definition).
+"""
+
+from . import _Experiment, register
+from .data import SyntheticClassification
+from .. import tools
+from ..models import RESNETS
+
+_DATASETS = {
+    "imagenet": {"shape": (3, 224, 224), "classes": 1000},
+    "cifar10": {"shape": (3, 32, 32), "classes": 10},
+}
+
+
+class ResNetExperiment(_Experiment):
+    def __init__(self, args, arch="resnet50", dataset="imagenet"):
+        args = tools.parse_keyval(args, defaults={
+            "batch-size": 32, "eval-batch-size": 256, "seed": 1234,
+            "eval-examples": 512, "image-size": 0})
+        if args["batch-size"] <= 0:
+            raise tools.UserException("Cannot make batches of non-positive size")
+        self.args = args
+        self.arch = arch
+        spec = _DATASETS[dataset]
+        shape = spec["shape"]
+        if args["image-size"] > 0:
+            shape = (shape[0], args["image-size"], args["image-size"])
+        self.classes = spec["classes"]
+        self._synth = SyntheticClassification(
+            shape, self.classes, seed=args["seed"],
+            eval_examples=args["eval-examples"])
+
+    def model(self):
+        return RESNETS[self.arch](num_classes=self.classes)
+
+    def train_batch(self, worker, step, device):
+        return self._synth.batch(self.args["batch-size"], worker, step, device)
+
+    def eval_batches(self, device):
+        yield from self._synth.eval_batches(self.args["eval-batch-size"], device)
+
+
+def _make(arch, dataset):
+    class _Bound(ResNetExperiment):
+        def __init__(self, args):
+            super().__init__(args, arch=arch, dataset=dataset)
+    _Bound.__name__ = f"ResNet_{arch}_{dataset}"
+    return _Bound
+
+
+for _arch in RESNETS:
+    for _ds in _DATASETS:
+        cls = _make(_arch, _ds)
+        register(f"{_arch}-{_ds}", cls)
+        # Reference slim-name alias, e.g. slim-resnet_v1_50-imagenet
+        # (experiments/slims.py naming convention).
+        _depth = _arch.replace("resnet", "")
+        register(f"slim-resnet_v1_{_depth}-{_ds}", cls)

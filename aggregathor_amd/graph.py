@@ -1,0 +1,293 @@
+"""Training engine: the MI355X-native equivalent of the reference's
+``graph.Manager`` (/root/reference/graph.py:204-315).
+
+Where the reference built one big TF graph (per-worker replica losses,
+gradient flatten, GAR node, inflate + apply on the parameter server), this
+engine drives the same pipeline imperatively per step:
+
+  1. per local worker: micro-batch forward + backward, with ``param.grad``
+     bound as views into that worker's row of the local gradient matrix --
+     autograd writes the flattened gradient IN PLACE (zero-copy flatten,
+     replacing graph.py:144-168's reshape/concat),
+  2. real-Byzantine workers' rows replaced by the attack plugin,
+  3. RCCL all-gather of the [n, d] matrix over xGMI (WorkerGroup.gather),
+  4. optional lossy-channel injection (UDP-semantics reproduction),
+  5. GAR kernel, replicated deterministically on every rank,
+  6. ``param.grad`` rebound as views into the aggregated flat gradient
+     (zero-copy inflate, graph.py:182-199) + optimizer step.
+
+Learning-rate schedules (fixed / polynomial / exponential) and the optimizer
+set (adadelta / adagrad / adam / rmsprop / sgd) mirror graph.py:51-66 with
+the same sub-argument names and defaults.
+"""
+
+import math
+
+import torch
+
+from . import aggregators as aggregators_mod
+from . import attacks as attacks_mod
+from . import config, tools
+
+# ---------------------------------------------------------------------------- #
+# Learning-rate schedules (reference graph.py:51-57)
+
+learning_rates = {
+    "fixed": {
+        "args": {"initial-rate": config.default_learning_rate},
+        "fn": lambda step, a: a["initial-rate"],
+    },
+    "polynomial": {
+        "args": {"initial-rate": config.default_learning_rate,
+                 "end-rate": config.default_end_learning_rate,
+                 "decay-step": config.default_decay_step,
+                 "power": 1.},
+        # tf.train.polynomial_decay with cycle=False (step clamped).
+        "fn": lambda step, a: (
+            (a["initial-rate"] - a["end-rate"])
+            * (1 - min(step, a["decay-step"]) / a["decay-step"]) ** a["power"]
+            + a["end-rate"]),
+    },
+    "exponential": {
+        "args": {"initial-rate": config.default_learning_rate,
+                 "decay-step": config.default_decay_step,
+                 "decay-rate": config.default_decay_rate},
+        # tf.train.exponential_decay, staircase=False.
+        "fn": lambda step, a: a["initial-rate"] * a["decay-rate"] ** (step / a["decay-step"]),
+    },
+}
+
+# Optimizers (reference graph.py:58-66); defaults follow the reference's
+# TF-1 hyper-parameters (e.g. adadelta eps=1.0, rmsprop decay=0.9 eps=1e-10).
+optimizers = {
+    "adadelta": {
+        "args": {"adadelta-rho": 0.95, "opt-epsilon": 1.},
+        "fn": lambda params, lr, a: torch.optim.Adadelta(
+            params, lr=lr, rho=a["adadelta-rho"], eps=a["opt-epsilon"]),
+    },
+    "adagrad": {
+        "args": {"initial-accumulator-value": 0.1},
+        "fn": lambda params, lr, a: torch.optim.Adagrad(
+            params, lr=lr, initial_accumulator_value=a["initial-accumulator-value"]),
+    },
+    "adam": {
+        "args": {"adam-beta1": 0.9, "adam-beta2": 0.999},
+        "fn": lambda params, lr, a: torch.optim.Adam(
+            params, lr=lr, betas=(a["adam-beta1"], a["adam-beta2"])),
+    },
+    "rmsprop": {
+        "args": {},
+        "fn": lambda params, lr, a: torch.optim.RMSprop(
+            params, lr=lr, alpha=0.9, eps=1e-10),
+    },
+    "sgd": {
+        "args": {},
+        "fn": lambda params, lr, a: torch.optim.SGD(params, lr=lr),
+    },
+}
+
+
+def build_lr_schedule(name, args):
+    if name not in learning_rates:
+        raise tools.UserException(
+            f"Unknown learning rate {name!r}, expected one of: "
+            + ", ".join(sorted(learning_rates)))
+    spec = learning_rates[name]
+    parsed = tools.parse_keyval(args, defaults=spec["args"])
+    known = {k: parsed[k] for k in spec["args"]}
+    return lambda step: spec["fn"](step, known)
+
+
+def build_optimizer(name, args, params, lr):
+    if name not in optimizers:
+        raise tools.UserException(
+            f"Unknown optimizer {name!r}, expected one of: "
+            + ", ".join(sorted(optimizers)))
+    spec = optimizers[name]
+    parsed = tools.parse_keyval(args, defaults=spec["args"])
+    known = {k: parsed[k] for k in spec["args"]}
+    return spec["fn"](params, lr, known)
+
+
+# ---------------------------------------------------------------------------- #
+# Flatten/inflate helpers (reference graph.py:144-199), view-based.
+
+
+def flat_size(params):
+    return sum(p.numel() for p in params)
+
+
+def bind_grad_views(params, flat):
+    """Bind each param's ``.grad`` as a view into the flat [d] buffer."""
+    off = 0
+    for p in params:
+        n = p.numel()
+        p.grad = flat[off:off + n].view_as(p)
+        off += n
+    assert off == flat.numel()
+
+
+# ---------------------------------------------------------------------------- #
+# Engine
+
+
+class Engine:
+    """One rank's training engine (all ranks run the same engine)."""
+
+    def __init__(self, experiment, aggregator, group,
+                 nbbyzwrks=0, aggregator_args=None,
+                 optimizer="sgd", optimizer_args=None,
+                 learning_rate="fixed", learning_rate_args=None,
+                 l1_regularize=-1., l2_regularize=-1.,
+                 nb_real_byz=0, attack=None, attack_args=None,
+                 lossy=None, amp=False, trace=False, seed=1234):
+        """
+        Args:
+          experiment: an instantiated _Experiment
+          aggregator: GAR name (see aggregathor_amd.aggregators.itemize())
+          group:      WorkerGroup (defines n and this rank's worker ids)
+          nbbyzwrks:  declared f for the GAR
+          nb_real_byz / attack / attack_args: real Byzantine workers
+                      (ids 0..nb_real_byz-1) and the attack they mount
+          lossy:      optional attacks.lossy.LossyChannel
+          amp:        bf16 autocast for forward/backward (fp32 gradients)
+        """
+        self.experiment = experiment
+        self.group = group
+        self.device = group.device
+        self.n = group.nbworkers
+        self.f = nbbyzwrks
+        self.amp = amp and self.device.type == "cuda"
+        self.trace = trace
+        self.lossy = lossy
+        self.global_step = 0
+
+        torch.manual_seed(seed)
+        self.model = experiment.model().to(self.device)
+        if self.device.type == "cuda":
+            self.model = self.model.to(memory_format=torch.channels_last) \
+                if any(p.dim() == 4 for p in self.model.parameters()) else self.model
+        group.broadcast_model(self.model)
+        self.params = [p for p in self.model.parameters() if p.requires_grad]
+        self.d = flat_size(self.params)
+
+        self.gar = aggregators_mod.instantiate(
+            aggregator, self.n, nbbyzwrks, aggregator_args or [])
+        self.l1 = l1_regularize
+        self.l2 = l2_regularize
+        self.lr_fn = build_lr_schedule(learning_rate, learning_rate_args or [])
+        self.optimizer = build_optimizer(
+            optimizer, optimizer_args or [], self.params, self.lr_fn(0))
+
+        self.nb_real_byz = nb_real_byz
+        self.attack = None
+        if nb_real_byz > 0 and attack:
+            self.attack = attacks_mod.instantiate(attack, attack_args or [])
+
+        # Resident buffers: local gradient rows, gathered matrix, aggregate.
+        lw = group.local_workers
+        self.local_rows = torch.zeros((lw, self.d), dtype=torch.float32,
+                                      device=self.device)
+        if group.distributed:
+            self.matrix = torch.empty((self.n, self.d), dtype=torch.float32,
+                                      device=self.device)
+        else:
+            self.matrix = self.local_rows
+        self.agg_flat = torch.zeros(self.d, dtype=torch.float32,
+                                    device=self.device)
+        self.last_loss = float("nan")
+
+    # ------------------------------------------------------------------ #
+
+    def _trace(self, msg):
+        if self.trace:
+            tools.trace(f"[step {self.global_step}] {msg}")
+
+    def _regularization(self):
+        """Reference graph.py:125-139: l1 = sum |w|; l2 = sqrt(sum w^2)."""
+        reg = None
+        if self.l1 > 0:
+            reg = self.l1 * sum(p.abs().sum() for p in self.params)
+        if self.l2 > 0:
+            l2 = self.l2 * torch.sqrt(sum((p * p).sum() for p in self.params))
+            reg = l2 if reg is None else reg + l2
+        return reg
+
+    def compute_local_gradients(self):
+        """Fill ``local_rows`` with this rank's worker gradients; returns the
+        mean loss over the local workers."""
+        self.model.train()
+        losses = []
+        for li, worker in enumerate(self.group.worker_ids):
+            row = self.local_rows[li]
+            row.zero_()
+            bind_grad_views(self.params, row)
+            batch = self.experiment.train_batch(worker, self.global_step, self.device)
+            self._trace(f"worker {worker}: forward")
+            if self.amp:
+                with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+                    loss = self.experiment.loss(self.model, batch)
+            else:
+                loss = self.experiment.loss(self.model, batch)
+            reg = self._regularization()
+            if reg is not None:
+                loss = loss + reg
+            self._trace(f"worker {worker}: backward")
+            loss.backward()
+            losses.append(loss.detach())
+            # Byzantine replacement: the worker computed honestly, then lies.
+            if self.attack is not None and worker < self.nb_real_byz:
+                self._trace(f"worker {worker}: byzantine craft")
+                row.copy_(self.attack.craft(row.clone(), worker, self.global_step))
+        return torch.stack(losses).mean()
+
+    def aggregate(self):
+        """Gather all rows, inject channel loss, run the GAR."""
+        self._trace("gather")
+        matrix = self.group.gather(self.local_rows, out=self.matrix if self.group.distributed else None)
+        if self.lossy is not None:
+            self._trace("lossy inject")
+            matrix = self.lossy.inject(matrix, self.global_step)
+        self._trace(f"aggregate ({type(self.gar).__name__})")
+        return self.gar.aggregate(matrix)
+
+    def apply(self, aggregated):
+        """Apply the aggregated gradient (identical on every rank)."""
+        self._trace("apply")
+        self.agg_flat.copy_(aggregated)
+        bind_grad_views(self.params, self.agg_flat)
+        lr = self.lr_fn(self.global_step)
+        for pg in self.optimizer.param_groups:
+            pg["lr"] = lr
+        self.optimizer.step()
+        self.global_step += 1
+
+    def step(self):
+        """One full training step; returns the local mean worker loss."""
+        loss = self.compute_local_gradients()
+        aggregated = self.aggregate()
+        self.apply(aggregated)
+        self.last_loss = loss.item()
+        return self.last_loss
+
+    # ------------------------------------------------------------------ #
+
+    def evaluate(self):
+        """Top-1 accuracy on the experiment's eval set (rank-0 buffers)."""
+        if self.group.distributed:
+            import torch.distributed as dist
+            for b in self.model.buffers():
+                dist.broadcast(b.data, src=0)
+        return self.experiment.accuracy(self.model, self.device)
+
+    def state_dict(self):
+        return {
+            "step": self.global_step,
+            "model": self.model.state_dict(),
+            "optimizer": self.optimizer.state_dict(),
+        }
+
+    def load_state_dict(self, state):
+        self.global_step = state["step"]
+        self.model.load_state_dict(state["model"])
+        self.optimizer.load_state_dict(state["optimizer"])
