@@ -1,0 +1,785 @@
+// gfx950 (CDNA4 / MI355X) kernels for the Byzantine-resilient GARs.
+//
+// Written MI355X-first from the algorithm definitions in the reference
+// (semantics of /root/reference/native/op_krum/cpu.cpp,
+// op_bulyan/cpu.cpp and aggregators/deprecated_native/native.cpp -- see
+// aggregathor_amd/ops/reference.py for the line-by-line mapping), NOT a
+// port of its CPU threadpool code:
+//
+// - The [n, d] gradient matrix (n <= 64 workers, d up to ~100M fp32) lives in
+//   HBM3E; every kernel is memory-bound, so the design minimizes passes over
+//   the matrix and vectorizes all global accesses as float4 (16 B/lane).
+// - Pairwise distances: one pass over the matrix; each block accumulates ALL
+//   pair partial sums for its d-chunk in registers (each loaded element is
+//   reused n-1 times), then a deterministic fixed-shape tree reduction
+//   produces the [n, n] matrix. No atomics -> bit-identical results on every
+//   rank, which is what lets the GAR run replicated instead of
+//   PS-and-broadcast.
+// - Krum / Bulyan selection: a single workgroup stages the n x n distance
+//   matrix in LDS and computes scores + top-k by rank counting (total order:
+//   non-finite last, then value, then index -- the reference's isfinite
+//   comparators refined to a deterministic tie-break). No host round trip:
+//   the selection result stays on device, so a whole GAR step is
+//   graph-capturable.
+// - Bulyan tail is FUSED: selection averages of t rounds + coordinate-wise
+//   averaged-median in ONE pass over [n, d] (the reference materialized a
+//   t x d intermediate and re-read it, op_bulyan/cpu.cpp:59-60,163-187).
+// - Coordinate-wise rules keep each column of n values in registers (rank
+//   selection, fully unrolled over a templated NMAX) -- columns are read
+//   coalesced since consecutive lanes hold consecutive coordinates.
+//
+// Numerics: fp32 throughout (the reference kernels are float/double; the
+// training path feeds fp32 flattened gradients). Compiled WITHOUT fast-math:
+// NaN ordering is load-bearing (NaN == +inf in every sort, NaN-skip in
+// means).
+
+#include <float.h>
+#include <hip/hip_runtime.h>
+
+#include "gar_kernels.h"
+
+namespace gar {
+
+constexpr int kBlock = 256;
+constexpr int kMaxBlocksD = 2048;  // cap + grid-stride (guideline 11)
+
+static inline int nblocks_d(long d) {
+  long d4 = d / 4;
+  long want = (d4 + kBlock - 1) / kBlock;
+  if (want < 1) want = 1;
+  if (want > kMaxBlocksD) want = kMaxBlocksD;
+  return (int)want;
+}
+
+// Aligned fp32 vector: rows of the [n, d] matrix start at arbitrary
+// d-multiples, so the usable vector width is the largest power of two
+// dividing d (a float4 access on a row with d % 4 != 0 would be
+// misaligned). Launchers dispatch VW in {4, 2, 1}.
+template <int VW>
+struct alignas(VW * 4) fvec {
+  float v[VW];
+};
+
+static inline int vec_width(long d) {
+  if (d % 4 == 0) return 4;
+  if (d % 2 == 0) return 2;
+  return 1;
+}
+
+// Pair enumeration layout of the partials buffer: the templated small-n
+// kernels enumerate pairs over NR compile-time rows; the generic tile kernel
+// enumerates over n directly. pair_index(i, j, rows) for i < j.
+__host__ __device__ static inline int pair_index(int i, int j, int rows) {
+  return i * (2 * rows - i - 1) / 2 + (j - i - 1);
+}
+
+static inline int pair_layout_rows(int n) {
+  if (n <= 8) return 8;
+  if (n <= 16) return 16;
+  return n;  // tile kernel uses n-enumeration
+}
+
+long sqdist_partials_elems(int n, long d) {
+  int rows = pair_layout_rows(n);
+  return (long)nblocks_d(d) * (rows * (rows - 1) / 2);
+}
+
+// ---------------------------------------------------------------------------
+// Total order: non-finite last, then value, then index. This is the
+// reference's isfinite comparator (op_krum/cpu.cpp:81-89) refined with a
+// deterministic index tie-break (matches the stable argsort of
+// ops/reference.py).
+__device__ __forceinline__ bool lt_total(float av, int ai, float bv, int bi) {
+  bool fa = isfinite(av), fb = isfinite(bv);
+  if (fa != fb) return fa;
+  if (!fa) return ai < bi;
+  if (av != bv) return av < bv;
+  return ai < bi;
+}
+
+// ---------------------------------------------------------------------------
+// Pairwise squared-L2 distances, stage 1: per-block partial sums.
+//
+// Small-n kernel (n <= NR, NR in {8, 16}): every block walks its grid-stride
+// share of the d dimension with float4 loads and keeps ALL NR*(NR-1)/2 pair
+// accumulators in VGPRs (NR=8: 28 acc + 8 float4 = ~75 VGPR; NR=16 uses
+// float2 loads to stay under the register cliff).
+
+template <int NR, int VW>
+__global__ __launch_bounds__(kBlock) void sqdist_small_kernel(
+    const float* __restrict__ g, float* __restrict__ partials, long dv,
+    long d, int n) {
+  constexpr int P = NR * (NR - 1) / 2;
+  float acc[P];
+#pragma unroll
+  for (int p = 0; p < P; ++p) acc[p] = 0.f;
+
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long x = (long)blockIdx.x * blockDim.x + threadIdx.x; x < dv;
+       x += stride) {
+    fvec<VW> v[NR];
+#pragma unroll
+    for (int i = 0; i < NR; ++i) {
+      if (i < n) {
+        v[i] = reinterpret_cast<const fvec<VW>*>(g + (long)i * d)[x];
+      } else {
+#pragma unroll
+        for (int c = 0; c < VW; ++c) v[i].v[c] = 0.f;
+      }
+    }
+    int p = 0;
+#pragma unroll
+    for (int i = 0; i < NR - 1; ++i) {
+#pragma unroll
+      for (int j = i + 1; j < NR; ++j, ++p) {
+        float s = 0.f;
+#pragma unroll
+        for (int c = 0; c < VW; ++c) {
+          float dd = v[i].v[c] - v[j].v[c];
+          s = fmaf(dd, dd, s);
+        }
+        acc[p] += s;
+      }
+    }
+  }
+  // Tail coordinates (d % VW), handled once by block 0.
+  long tail0 = dv * VW;
+  if (blockIdx.x == 0 && threadIdx.x < (int)(d - tail0)) {
+    long x = tail0 + threadIdx.x;
+    int p = 0;
+#pragma unroll
+    for (int i = 0; i < NR - 1; ++i) {
+#pragma unroll
+      for (int j = i + 1; j < NR; ++j, ++p) {
+        if (j < n) {
+          float dd = g[(long)i * d + x] - g[(long)j * d + x];
+          acc[p] = fmaf(dd, dd, acc[p]);
+        }
+      }
+    }
+  }
+  // Deterministic per-pair block reduction (fixed tree shape).
+  __shared__ float red[kBlock];
+  for (int p = 0; p < P; ++p) {
+    red[threadIdx.x] = acc[p];
+    __syncthreads();
+    for (int s = kBlock / 2; s > 0; s >>= 1) {
+      if ((int)threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[(long)blockIdx.x * P + p] = red[0];
+    __syncthreads();
+  }
+}
+
+// Generic row-tile kernel for n > 16: blockIdx.y selects an 8x8 tile of the
+// upper-triangular pair space; each row chunk is re-read ~n/8 times (still
+// one HBM pass per tile thanks to L2), accumulators stay at 64 VGPRs.
+template <int VW>
+__global__ __launch_bounds__(kBlock) void sqdist_tile_kernel(
+    const float* __restrict__ g, float* __restrict__ partials, long dv,
+    long d, int n, int tiles_per_row) {
+  // Decode upper-triangular tile (ti <= tj).
+  int tile = blockIdx.y;
+  int ti = 0;
+  {
+    int rem = tile;
+    while (rem >= tiles_per_row - ti) {
+      rem -= tiles_per_row - ti;
+      ++ti;
+    }
+    tile = rem;
+  }
+  int tj = ti + tile;
+  int i0 = ti * 8, j0 = tj * 8;
+  const int P = n * (n - 1) / 2;
+
+  float acc[64];
+#pragma unroll
+  for (int p = 0; p < 64; ++p) acc[p] = 0.f;
+
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long x = (long)blockIdx.x * blockDim.x + threadIdx.x; x < dv;
+       x += stride) {
+    fvec<VW> vi[8], vj[8];
+#pragma unroll
+    for (int a = 0; a < 8; ++a) {
+      int gi = i0 + a, gj = j0 + a;
+      if (gi < n) {
+        vi[a] = reinterpret_cast<const fvec<VW>*>(g + (long)gi * d)[x];
+      } else {
+#pragma unroll
+        for (int c = 0; c < VW; ++c) vi[a].v[c] = 0.f;
+      }
+      if (gj < n) {
+        vj[a] = reinterpret_cast<const fvec<VW>*>(g + (long)gj * d)[x];
+      } else {
+#pragma unroll
+        for (int c = 0; c < VW; ++c) vj[a].v[c] = 0.f;
+      }
+    }
+#pragma unroll
+    for (int a = 0; a < 8; ++a) {
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        float s = 0.f;
+#pragma unroll
+        for (int c = 0; c < VW; ++c) {
+          float dd = vi[a].v[c] - vj[b].v[c];
+          s = fmaf(dd, dd, s);
+        }
+        acc[a * 8 + b] += s;
+      }
+    }
+  }
+  long tail0 = dv * VW;
+  if (blockIdx.x == 0 && threadIdx.x < (int)(d - tail0)) {
+    long x = tail0 + threadIdx.x;
+#pragma unroll
+    for (int a = 0; a < 8; ++a) {
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        int gi = i0 + a, gj = j0 + b;
+        if (gi < n && gj < n && gi < gj) {
+          float dd = g[(long)gi * d + x] - g[(long)gj * d + x];
+          acc[a * 8 + b] = fmaf(dd, dd, acc[a * 8 + b]);
+        }
+      }
+    }
+  }
+  __shared__ float red[kBlock];
+  for (int a = 0; a < 8; ++a) {
+    for (int b = 0; b < 8; ++b) {
+      int gi = i0 + a, gj = j0 + b;
+      if (!(gi < n && gj < n && gi < gj)) continue;
+      red[threadIdx.x] = acc[a * 8 + b];
+      __syncthreads();
+      for (int s = kBlock / 2; s > 0; s >>= 1) {
+        if ((int)threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+        __syncthreads();
+      }
+      if (threadIdx.x == 0)
+        partials[(long)blockIdx.x * P + pair_index(gi, gj, n)] = red[0];
+      __syncthreads();
+    }
+  }
+}
+
+// Stage 2: deterministic cross-block reduction -> [n, n] symmetric matrix
+// with +inf diagonal (the diagonal is never a candidate; the reference
+// stores T::max there, op_bulyan/cpu.cpp:75).
+__global__ void sqdist_reduce_kernel(const float* __restrict__ partials,
+                                     float* __restrict__ dist, int nblk,
+                                     int n, int layout_rows) {
+  int p = blockIdx.x;  // n-enumeration pair index
+  // Decode p -> (i, j).
+  int i = 0, rem = p;
+  while (rem >= n - 1 - i) {
+    rem -= n - 1 - i;
+    ++i;
+  }
+  int j = i + 1 + rem;
+  int P = layout_rows * (layout_rows - 1) / 2;
+  int pl = pair_index(i, j, layout_rows);
+
+  __shared__ float red[kBlock];
+  float s = 0.f;
+  for (int b = threadIdx.x; b < nblk; b += kBlock)
+    s += partials[(long)b * P + pl];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int st = kBlock / 2; st > 0; st >>= 1) {
+    if ((int)threadIdx.x < st) red[threadIdx.x] += red[threadIdx.x + st];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    dist[(long)i * n + j] = red[0];
+    dist[(long)j * n + i] = red[0];
+  }
+}
+
+__global__ void fill_diag_inf_kernel(float* dist, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dist[(long)i * n + i] = HUGE_VALF;
+}
+
+void sqdist(const float* g, int n, long d, float* partials, float* dist,
+            hipStream_t stream) {
+  int nblk = nblocks_d(d);
+  int vw = vec_width(d);
+  if (n <= 8) {
+    if (vw == 4)
+      sqdist_small_kernel<8, 4>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d / 4, d, n);
+    else if (vw == 2)
+      sqdist_small_kernel<8, 2>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d / 2, d, n);
+    else
+      sqdist_small_kernel<8, 1>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
+  } else if (n <= 16) {
+    if (vw >= 2)
+      sqdist_small_kernel<16, 2>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d / 2, d, n);
+    else
+      sqdist_small_kernel<16, 1>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
+  } else {
+    int tiles_per_row = (n + 7) / 8;
+    int ntiles = tiles_per_row * (tiles_per_row + 1) / 2;
+    dim3 grid(nblk, ntiles);
+    if (vw >= 2)
+      sqdist_tile_kernel<2><<<grid, kBlock, 0, stream>>>(g, partials, d / 2,
+                                                         d, n, tiles_per_row);
+    else
+      sqdist_tile_kernel<1><<<grid, kBlock, 0, stream>>>(g, partials, d, d,
+                                                         n, tiles_per_row);
+  }
+  int P = n * (n - 1) / 2;
+  sqdist_reduce_kernel<<<P, kBlock, 0, stream>>>(partials, dist, nblk, n,
+                                                 pair_layout_rows(n));
+  fill_diag_inf_kernel<<<1, 64, 0, stream>>>(dist, n);
+}
+
+// ---------------------------------------------------------------------------
+// Multi-Krum selection (one workgroup, LDS-staged; op_krum/cpu.cpp:74-119
+// semantics). Scores and top-m by rank counting under the total order.
+
+__global__ void krum_select_kernel(const float* __restrict__ dist, int n,
+                                   int f, int m, int* __restrict__ sel) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sd = reinterpret_cast<float*>(smem);  // n*n
+  float* scores = sd + n * n;                  // n
+  for (int i = threadIdx.x; i < n * n; i += blockDim.x) sd[i] = dist[i];
+  __syncthreads();
+  const int nbin = n - f - 2;
+  const int i = threadIdx.x;
+  if (i < n) {
+    // score_i = sum of i's nbin smallest distances (rank < nbin under the
+    // total order == stable-sorted prefix).
+    float score = 0.f;
+    for (int j = 0; j < n; ++j) {
+      if (j == i) continue;
+      float vj = sd[i * n + j];
+      int rank = 0;
+      for (int l = 0; l < n; ++l) {
+        if (l == i || l == j) continue;
+        if (lt_total(sd[i * n + l], l, vj, j)) ++rank;
+      }
+      if (rank < nbin) score += vj;
+    }
+    scores[i] = score;
+  }
+  __syncthreads();
+  if (i < n) {
+    float vi = scores[i];
+    int rank = 0;
+    for (int l = 0; l < n; ++l)
+      if (l != i && lt_total(scores[l], l, vi, i)) ++rank;
+    if (rank < m) sel[rank] = i;  // ascending-score order by construction
+  }
+}
+
+void krum_select(const float* dist, int n, int f, int m, int* sel,
+                 hipStream_t stream) {
+  size_t lds = (size_t)(n * n + n) * sizeof(float);
+  krum_select_kernel<<<1, kBlock, lds, stream>>>(dist, n, f, m, sel);
+}
+
+// ---------------------------------------------------------------------------
+// Selection average: out = mean of the m selected rows (operations.hpp:67-76).
+
+template <int VW>
+__global__ __launch_bounds__(kBlock) void selection_average_kernel(
+    const float* __restrict__ g, float* __restrict__ out, long dv, long d,
+    const int* __restrict__ sel_g, int m) {
+  __shared__ int sel[kMaxNSelect];
+  if ((int)threadIdx.x < m) sel[threadIdx.x] = sel_g[threadIdx.x];
+  __syncthreads();
+  const float inv = 1.f / (float)m;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long x = (long)blockIdx.x * blockDim.x + threadIdx.x; x < dv;
+       x += stride) {
+    fvec<VW> s;
+#pragma unroll
+    for (int c = 0; c < VW; ++c) s.v[c] = 0.f;
+    for (int i = 0; i < m; ++i) {
+      fvec<VW> v = reinterpret_cast<const fvec<VW>*>(g + (long)sel[i] * d)[x];
+#pragma unroll
+      for (int c = 0; c < VW; ++c) s.v[c] += v.v[c];
+    }
+#pragma unroll
+    for (int c = 0; c < VW; ++c) s.v[c] *= inv;
+    reinterpret_cast<fvec<VW>*>(out)[x] = s;
+  }
+  long tail0 = dv * VW;
+  if (blockIdx.x == 0 && threadIdx.x < (int)(d - tail0)) {
+    long x = tail0 + threadIdx.x;
+    float s = 0.f;
+    for (int i = 0; i < m; ++i) s += g[(long)sel[i] * d + x];
+    out[x] = s * inv;
+  }
+}
+
+void selection_average(const float* g, int n, long d, const int* sel, int m,
+                       float* out, hipStream_t stream) {
+  (void)n;
+  int nblk = nblocks_d(d);
+  int vw = vec_width(d);
+  if (vw == 4)
+    selection_average_kernel<4>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d / 4, d, sel, m);
+  else if (vw == 2)
+    selection_average_kernel<2>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d / 2, d, sel, m);
+  else
+    selection_average_kernel<1>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d, d, sel, m);
+}
+
+// ---------------------------------------------------------------------------
+// Bulyan selection schedule (one workgroup; op_bulyan/cpu.cpp:88-161
+// semantics: initial scores + per-row distance pruning, then t rounds of
+// select-average-evict with score decrement by the pruned distance).
+
+__global__ void bulyan_select_kernel(const float* __restrict__ dist, int n,
+                                     int f, int m,
+                                     unsigned char* __restrict__ flags) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sd = reinterpret_cast<float*>(smem);  // n*n, pruned in place
+  float* scores = sd + n * n;                  // n
+  int* evicted = reinterpret_cast<int*>(scores + n);
+  unsigned char* alive = reinterpret_cast<unsigned char*>(evicted + 1);
+  for (int i = threadIdx.x; i < n * n; i += blockDim.x) sd[i] = dist[i];
+  __syncthreads();
+  const int nbin = n - f - 2;
+  const int t = n - 2 * f - 2;
+  const int i = threadIdx.x;
+  if (i < n) {
+    float score = 0.f;
+    unsigned long long keep = 0ull;  // n <= 64
+    for (int j = 0; j < n; ++j) {
+      if (j == i) continue;
+      float vj = sd[i * n + j];
+      int rank = 0;
+      for (int l = 0; l < n; ++l) {
+        if (l == i || l == j) continue;
+        if (lt_total(sd[i * n + l], l, vj, j)) ++rank;
+      }
+      if (rank < nbin) {
+        score += vj;
+        keep |= 1ull << j;
+      }
+    }
+    // Prune row i: non-kept distances (and the diagonal) become 0
+    // (cpu.cpp:116-129), so the eviction decrement reads 0 for them.
+    for (int j = 0; j < n; ++j)
+      if (j == i || !((keep >> j) & 1ull)) sd[i * n + j] = 0.f;
+    scores[i] = score;
+    alive[i] = 1;
+  }
+  __syncthreads();
+  for (int k = 0;; ++k) {
+    if (i < n) {
+      float vi = scores[i];
+      int rank = 0;
+      for (int l = 0; l < n; ++l)
+        if (l != i && lt_total(scores[l], l, vi, i)) ++rank;
+      flags[(long)k * n + i] = (rank < m - k) ? 1 : 0;
+      if (rank == 0) *evicted = i;
+    }
+    __syncthreads();
+    if (k + 1 >= t) break;
+    int ev = *evicted;
+    if (i < n) {
+      if (i == ev) {
+        scores[i] = FLT_MAX;  // reference uses T::max (cpu.cpp:154)
+        alive[i] = 0;
+      } else if (alive[i]) {
+        scores[i] -= sd[i * n + ev];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+long bulyan_flags_bytes(int n, int f) {
+  const int t = n - 2 * f - 2;
+  return ((long)(t * n + 15) & ~15L) + (long)t * sizeof(float);
+}
+
+void bulyan_select(const float* dist, int n, int f, int m,
+                   unsigned char* flags, hipStream_t stream) {
+  size_t lds = (size_t)(n * n + n) * sizeof(float) + sizeof(int) + n + 16;
+  bulyan_select_kernel<<<1, kBlock, lds, stream>>>(dist, n, f, m, flags);
+}
+
+// ---------------------------------------------------------------------------
+// Per-coordinate rank-selection helpers (fully unrolled over NMAX so the
+// column stays in VGPRs; n is runtime <= NMAX).
+
+template <int NMAX>
+__device__ __forceinline__ float coord_median(const float (&vals)[NMAX],
+                                              int n) {
+  // Element at rank n/2 under the total order (native.cpp:686-694).
+  const int target = n / 2;
+  float out = 0.f;
+#pragma unroll
+  for (int i = 0; i < NMAX; ++i) {
+    if (i >= n) break;
+    int rank = 0;
+#pragma unroll
+    for (int j = 0; j < NMAX; ++j) {
+      if (j >= n) break;
+      if (j != i && lt_total(vals[j], j, vals[i], i)) ++rank;
+    }
+    if (rank == target) out = vals[i];
+  }
+  return out;
+}
+
+template <int NMAX>
+__device__ __forceinline__ float coord_averaged_median(
+    const float (&vals)[NMAX], int n, int beta) {
+  // Mean of the beta elements closest to the median (native.cpp:714-739).
+  const float zero = coord_median<NMAX>(vals, n);
+  float delta[NMAX];
+#pragma unroll
+  for (int i = 0; i < NMAX; ++i) {
+    if (i >= n) break;
+    delta[i] = fabsf(vals[i] - zero);
+  }
+  float sum = 0.f;
+#pragma unroll
+  for (int i = 0; i < NMAX; ++i) {
+    if (i >= n) break;
+    int rank = 0;
+#pragma unroll
+    for (int j = 0; j < NMAX; ++j) {
+      if (j >= n) break;
+      if (j != i && lt_total(delta[j], j, delta[i], i)) ++rank;
+    }
+    if (rank < beta) sum += vals[i];
+  }
+  return sum / (float)beta;
+}
+
+// Load one float4-column block of the matrix into registers, then run a
+// scalar per-coordinate functor on each of the 4 components. OP signature:
+// float op(const float (&vals)[NMAX], int n).
+template <int NMAX, int VW, class OP>
+__global__ __launch_bounds__(kBlock) void coordwise_kernel(
+    const float* __restrict__ g, float* __restrict__ out, long dv, long d,
+    int n, OP op) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long x = (long)blockIdx.x * blockDim.x + threadIdx.x; x < dv;
+       x += stride) {
+    fvec<VW> colv[NMAX];
+#pragma unroll
+    for (int i = 0; i < NMAX; ++i) {
+      if (i >= n) break;
+      colv[i] = reinterpret_cast<const fvec<VW>*>(g + (long)i * d)[x];
+    }
+    fvec<VW> res;
+#pragma unroll
+    for (int c = 0; c < VW; ++c) {
+      float vals[NMAX];
+#pragma unroll
+      for (int i = 0; i < NMAX; ++i) {
+        if (i >= n) break;
+        vals[i] = colv[i].v[c];
+      }
+      res.v[c] = op(vals, n);
+    }
+    reinterpret_cast<fvec<VW>*>(out)[x] = res;
+  }
+  long tail0 = dv * VW;
+  if (blockIdx.x == 0 && threadIdx.x < (int)(d - tail0)) {
+    long x = tail0 + threadIdx.x;
+    float vals[NMAX];
+#pragma unroll
+    for (int i = 0; i < NMAX; ++i) {
+      if (i >= n) break;
+      vals[i] = g[(long)i * d + x];
+    }
+    out[x] = op(vals, n);
+  }
+}
+
+template <int NMAX>
+struct MedianOp {
+  __device__ float operator()(const float (&vals)[NMAX], int n) const {
+    return coord_median<NMAX>(vals, n);
+  }
+};
+
+template <int NMAX>
+struct AveragedMedianOp {
+  int beta;
+  __device__ float operator()(const float (&vals)[NMAX], int n) const {
+    return coord_averaged_median<NMAX>(vals, n, beta);
+  }
+};
+
+template <int NMAX>
+struct AverageNanOp {
+  // Mean of the finite values; 0/0 -> NaN (native.cpp:756-775).
+  __device__ float operator()(const float (&vals)[NMAX], int n) const {
+    float sum = 0.f;
+    float count = 0.f;
+#pragma unroll
+    for (int i = 0; i < NMAX; ++i) {
+      if (i >= n) break;
+      if (isfinite(vals[i])) {
+        sum += vals[i];
+        count += 1.f;
+      }
+    }
+    return sum / count;
+  }
+};
+
+template <int NMAX, template <int> class OP, class... Args>
+static void launch_coordwise_n(const float* g, int n, long d, float* out,
+                               hipStream_t stream, Args... args) {
+  int nblk = nblocks_d(d);
+  int vw = vec_width(d);
+  if (vw == 4)
+    coordwise_kernel<NMAX, 4, OP<NMAX>>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d / 4, d, n, OP<NMAX>{args...});
+  else if (vw == 2)
+    coordwise_kernel<NMAX, 2, OP<NMAX>>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d / 2, d, n, OP<NMAX>{args...});
+  else
+    coordwise_kernel<NMAX, 1, OP<NMAX>>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d, d, n, OP<NMAX>{args...});
+}
+
+template <template <int> class OP, class... Args>
+static void launch_coordwise(const float* g, int n, long d, float* out,
+                             hipStream_t stream, Args... args) {
+  if (n <= 8)
+    launch_coordwise_n<8, OP>(g, n, d, out, stream, args...);
+  else if (n <= 16)
+    launch_coordwise_n<16, OP>(g, n, d, out, stream, args...);
+  else
+    launch_coordwise_n<kMaxNCoord, OP>(g, n, d, out, stream, args...);
+}
+
+void median(const float* g, int n, long d, float* out, hipStream_t stream) {
+  launch_coordwise<MedianOp>(g, n, d, out, stream);
+}
+
+void averaged_median(const float* g, int n, long d, int beta, float* out,
+                     hipStream_t stream) {
+  launch_coordwise<AveragedMedianOp>(g, n, d, out, stream, beta);
+}
+
+void average_nan(const float* g, int n, long d, float* out,
+                 hipStream_t stream) {
+  launch_coordwise<AverageNanOp>(g, n, d, out, stream);
+}
+
+// ---------------------------------------------------------------------------
+// Fused Bulyan tail: per coordinate, compute the t selection averages (from
+// the flags schedule) and the averaged-median over them, in one pass over
+// [n, d] (the reference wrote a t x d intermediate, op_bulyan/cpu.cpp:59-60).
+
+template <int NMAX, int VW>
+__global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
+    const float* __restrict__ g, float* __restrict__ out, long dv, long d,
+    int n, int t, int b, const unsigned char* __restrict__ flags_g,
+    const float* __restrict__ inv_mk) {
+  __shared__ unsigned char fl[NMAX * NMAX];
+  __shared__ float inv[NMAX];
+  for (int i = threadIdx.x; i < t * n; i += blockDim.x) fl[i] = flags_g[i];
+  if ((int)threadIdx.x < t) inv[threadIdx.x] = inv_mk[threadIdx.x];
+  __syncthreads();
+
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long x = (long)blockIdx.x * blockDim.x + threadIdx.x; x < dv;
+       x += stride) {
+    fvec<VW> colv[NMAX];
+#pragma unroll
+    for (int i = 0; i < NMAX; ++i) {
+      if (i >= n) break;
+      colv[i] = reinterpret_cast<const fvec<VW>*>(g + (long)i * d)[x];
+    }
+    fvec<VW> res;
+#pragma unroll
+    for (int c = 0; c < VW; ++c) {
+      float inters[NMAX];
+#pragma unroll
+      for (int k = 0; k < NMAX; ++k) {
+        if (k >= t) break;
+        float s = 0.f;
+#pragma unroll
+        for (int i = 0; i < NMAX; ++i) {
+          if (i >= n) break;
+          if (fl[k * n + i]) s += colv[i].v[c];
+        }
+        inters[k] = s * inv[k];
+      }
+      res.v[c] = coord_averaged_median<NMAX>(inters, t, b);
+    }
+    reinterpret_cast<fvec<VW>*>(out)[x] = res;
+  }
+  long tail0 = dv * VW;
+  if (blockIdx.x == 0 && threadIdx.x < (int)(d - tail0)) {
+    long x = tail0 + threadIdx.x;
+    float inters[NMAX];
+#pragma unroll
+    for (int k = 0; k < NMAX; ++k) {
+      if (k >= t) break;
+      float s = 0.f;
+#pragma unroll
+      for (int i = 0; i < NMAX; ++i) {
+        if (i >= n) break;
+        if (fl[k * n + i]) s += g[(long)i * d + x];
+      }
+      inters[k] = s * inv[k];
+    }
+    out[x] = coord_averaged_median<NMAX>(inters, t, b);
+  }
+}
+
+__global__ void bulyan_inv_mk_kernel(float* inv_mk, int t, int m) {
+  int k = threadIdx.x;
+  if (k < t) inv_mk[k] = 1.f / (float)(m - k);
+}
+
+void bulyan_final(const float* g, int n, long d, int f, int m,
+                  const unsigned char* flags, float* out,
+                  hipStream_t stream) {
+  const int t = n - 2 * f - 2;
+  const int b = t - 2 * f;
+  // Small helper buffer for 1/(m-k): lives at the end of the flags buffer?
+  // Simpler: compute on the fly in a tiny kernel into a static buffer is not
+  // graph-safe; instead reuse the flags allocation convention: the caller
+  // allocates flags of t*n bytes PLUS t floats (aligned) right after.
+  float* inv_mk = reinterpret_cast<float*>(
+      const_cast<unsigned char*>(flags) + ((t * n + 15) & ~15));
+  bulyan_inv_mk_kernel<<<1, 64, 0, stream>>>(inv_mk, t, m);
+  int nblk = nblocks_d(d);
+  int vw = vec_width(d);
+  long dv = d / vw;
+#define GAR_BULYAN_LAUNCH(NMAX, VW)                                   \
+  bulyan_final_kernel<NMAX, VW><<<nblk, kBlock, 0, stream>>>(         \
+      g, out, dv, d, n, t, b, flags, inv_mk)
+  if (n <= 8) {
+    if (vw == 4) GAR_BULYAN_LAUNCH(8, 4);
+    else if (vw == 2) GAR_BULYAN_LAUNCH(8, 2);
+    else GAR_BULYAN_LAUNCH(8, 1);
+  } else if (n <= 16) {
+    if (vw == 4) GAR_BULYAN_LAUNCH(16, 4);
+    else if (vw == 2) GAR_BULYAN_LAUNCH(16, 2);
+    else GAR_BULYAN_LAUNCH(16, 1);
+  } else {
+    if (vw == 4) GAR_BULYAN_LAUNCH(kMaxNCoord, 4);
+    else if (vw == 2) GAR_BULYAN_LAUNCH(kMaxNCoord, 2);
+    else GAR_BULYAN_LAUNCH(kMaxNCoord, 1);
+  }
+#undef GAR_BULYAN_LAUNCH
+}
+
+}  // namespace gar
