@@ -1,0 +1,131 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50, n=8 workers, f=2, Krum (vs Bulyan/average).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+runs the BASELINE.json headline config -- steps/sec of Byzantine-resilient
+data-parallel ResNet-50 training with n=8 workers and f=2 -- on N GPUs of one
+node (one rank per GPU over RCCL when launched via torch.distributed.run;
+single process when N=1). The total worker count stays n=8 at every N (the
+GAR's n/f are properties of the training run, not the GPU count), so per-GPU
+work shrinks as N grows: STRONG scaling, fixed global batch.
+
+Synthetic ImageNet-shaped data (3x224x224, 1000 classes), random-init
+weights, bf16 autocast compute with fp32 gradients/aggregation/optimizer.
+
+Rank 0 prints exactly one JSON line with the whole-job steps/sec (max-over-
+ranks timing, barrier+synchronize bracketed).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", type=str, default="resnet50")
+    ap.add_argument("--workers", type=int, default=8, help="total GAR workers n")
+    ap.add_argument("--f", type=int, default=2, help="declared Byzantine workers")
+    ap.add_argument("--gar", type=str, default="krum",
+                    help="aggregation rule (krum | bulyan | average | ...)")
+    ap.add_argument("--batch-size", type=int, default=32, help="per-worker batch")
+    ap.add_argument("--image-size", type=int, default=224)
+    ap.add_argument("--attack", type=str, default="",
+                    help="optional attack (e.g. reversal) mounted by f workers")
+    ap.add_argument("--device", type=str, default="",
+                    help="override device (debug; cpu allowed)")
+    ap.add_argument("--no-amp", action="store_true",
+                    help="disable bf16 autocast (fp32 compute)")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world)
+
+    if args.device:
+        device = args.device
+    elif torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        print("[bench] WARNING: no GPU visible, running on CPU (debug only)",
+              file=sys.stderr)
+        device = "cpu"
+
+    from aggregathor_amd import experiments, ops
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+
+    if device.startswith("cuda") and not ops.hip_available():
+        raise RuntimeError("HIP extension _gar_hip not built -- run "
+                           "`python -m aggregathor_amd.ops.build` first")
+
+    exp = experiments.instantiate(
+        f"{args.model}-imagenet",
+        [f"batch-size:{args.batch_size}", f"image-size:{args.image_size}",
+         "eval-examples:0"])
+    group = WorkerGroup(args.workers, device=device)
+    amp = not args.no_amp
+    engine = Engine(
+        exp, args.gar, group, nbbyzwrks=args.f, amp=amp,
+        nb_real_byz=(args.f if args.attack else 0), attack=args.attack,
+        optimizer="sgd", learning_rate="fixed")
+
+    def sync():
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+        group.barrier()
+
+    for _ in range(args.warmup):
+        engine.step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        engine.step()
+    sync()
+    elapsed = time.perf_counter() - t0
+    elapsed = group.allreduce_max(elapsed)  # worst rank defines job time
+
+    steps_per_sec = args.steps / elapsed
+    global_batch = args.batch_size * args.workers
+    if rank == 0:
+        result = {
+            "metric": f"steps/sec {args.model} n={args.workers} f={args.f} "
+                      f"{args.gar}",
+            "value": steps_per_sec,
+            "unit": "steps/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,  # reference publishes no in-tree numbers
+            "dtype": "bf16" if amp else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": args.image_size,
+                "parallelism": f"dp{n_gpus}",
+                "gar": args.gar,
+                "n_workers": args.workers,
+                "f": args.f,
+                "attack": args.attack or None,
+                "images_per_sec": steps_per_sec * global_batch,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()
