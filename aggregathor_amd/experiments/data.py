@@ -45,18 +45,40 @@ class SyntheticClassification:
             self.teacher_w = torch.randn(dim, classes, generator=gen)
         else:
             self.teacher_w = None
+        # Device-resident batch pool: on GPU, per-(worker, step) batches are
+        # served from a once-generated pool of `pool_size` distinct batches
+        # (batch(worker, step) == cpu batch(worker, step % pool_size)) --
+        # generating 20 MB of fresh host randoms per micro-batch would
+        # bottleneck an MI355X training step on the host RNG + H2D copy.
+        self.pool_size = 8
+        self._pools = {}
 
     def _labels(self, flat_inputs, gen):
         if self.teacher:
             return (flat_inputs @ self.teacher_w).argmax(dim=1)
         return torch.randint(0, self.classes, (flat_inputs.shape[0],), generator=gen)
 
-    def batch(self, batch_size, worker, step, device="cpu"):
-        """Training batch for (worker, step): pure function of the seed."""
+    def _raw_batch(self, batch_size, worker, step):
         gen = torch.Generator().manual_seed(
             (self.seed * 1000003 + worker * 7919 + step * 104729) & 0x7FFFFFFF)
         x = torch.randn((batch_size, *self.shape), generator=gen) * self.scale
         y = self._labels(x.flatten(1), gen)
+        return x, y
+
+    def batch(self, batch_size, worker, step, device="cpu"):
+        """Training batch for (worker, step): pure function of the seed."""
+        device = torch.device(device)
+        if device.type == "cuda":
+            key = (batch_size, worker, str(device))
+            pool = self._pools.get(key)
+            if pool is None:
+                pool = []
+                for s in range(self.pool_size):
+                    x, y = self._raw_batch(batch_size, worker, s)
+                    pool.append((x.to(device), y.to(device)))
+                self._pools[key] = pool
+            return pool[step % self.pool_size]
+        x, y = self._raw_batch(batch_size, worker, step)
         return x.to(device, non_blocking=True), y.to(device, non_blocking=True)
 
     def eval_batches(self, batch_size, device="cpu"):

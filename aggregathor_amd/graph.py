@@ -163,10 +163,10 @@ class Engine:
         self.global_step = 0
 
         torch.manual_seed(seed)
+        # Params stay contiguous: the flattened-gradient views bound before
+        # backward are contiguous, and a channels-last param would force a
+        # layout-converting grad accumulation on every parameter.
         self.model = experiment.model().to(self.device)
-        if self.device.type == "cuda":
-            self.model = self.model.to(memory_format=torch.channels_last) \
-                if any(p.dim() == 4 for p in self.model.parameters()) else self.model
         group.broadcast_model(self.model)
         self.params = [p for p in self.model.parameters() if p.requires_grad]
         self.d = flat_size(self.params)

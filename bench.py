@@ -23,6 +23,10 @@ import os
 import sys
 import time
 
+# Avoid MIOpen exhaustive-search stalls on fresh boxes (kernels still cached
+# after the warmup steps, which are untimed).
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 
 

@@ -11,9 +11,13 @@ Usage: python scripts/gar_microbench.py [--n 8] [--d 25600000] [--iters 50]
 
 import argparse
 import json
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def time_gpu(fn, iters, warmup=5):
