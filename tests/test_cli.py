@@ -1,0 +1,106 @@
+"""CLI integration: runner.py flag surface, checkpoint/eval layout, resume,
+deploy.py local multi-process launch."""
+
+import json
+import os
+import pathlib
+import subprocess
+import sys
+
+import pytest
+
+REPO = pathlib.Path(__file__).resolve().parent.parent
+
+
+def _run(cmd, timeout=300, env_extra=None):
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE", "LOCAL_RANK",
+                        "MASTER_ADDR", "MASTER_PORT")}
+    if env_extra:
+        env.update(env_extra)
+    return subprocess.run([sys.executable] + cmd, capture_output=True,
+                          timeout=timeout, env=env, cwd=str(REPO))
+
+
+def test_runner_mnist_average(tmp_path):
+    ckpt = tmp_path / "ckpt"
+    r = _run(["runner.py", "--experiment", "mnist", "--aggregator", "average",
+              "--nb-workers", "3", "--max-step", "12",
+              "--experiment-args", "batch-size:16",
+              "--checkpoint-dir", str(ckpt),
+              "--checkpoint-delta", "5", "--checkpoint-period", "-1",
+              "--evaluation-delta", "6", "--evaluation-period", "-1",
+              "--summary-delta", "6", "--summary-period", "-1",
+              "--progress-every", "5"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    out = r.stdout.decode()
+    assert "Steps/s" in out
+    # Checkpoint layout: model-<step>.ckpt files.
+    files = sorted(ckpt.glob("model-*.ckpt"))
+    assert files, "no checkpoints written"
+    # Eval TSV: wall-time \t step \t top1-X-acc:value
+    eval_file = ckpt / "eval"
+    assert eval_file.exists()
+    line = eval_file.read_text().strip().split("\n")[0].split("\t")
+    assert len(line) >= 3 and line[2].startswith("top1-X-acc:")
+    # Summary JSONL
+    recs = [json.loads(l) for l in (ckpt / "summary.jsonl").read_text().splitlines()]
+    assert all("loss" in r and "lr" in r for r in recs)
+
+
+def test_runner_resume(tmp_path):
+    ckpt = tmp_path / "ckpt"
+    base = ["runner.py", "--experiment", "mnist", "--aggregator", "krum",
+            "--nb-workers", "5", "--nb-decl-byz-workers", "1",
+            "--experiment-args", "batch-size:16",
+            "--checkpoint-dir", str(ckpt),
+            "--checkpoint-delta", "4", "--checkpoint-period", "-1",
+            "--evaluation-delta", "-1", "--evaluation-period", "-1",
+            "--progress-every", "0"]
+    r = _run(base + ["--max-step", "8"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    r = _run(base + ["--max-step", "14"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    assert "Restored checkpoint" in r.stdout.decode()
+    steps = sorted(int(p.stem.split("-")[1]) for p in ckpt.glob("model-*.ckpt"))
+    assert steps[-1] >= 12  # continued past the first run's 8
+
+
+def test_runner_attack_flags(tmp_path):
+    r = _run(["runner.py", "--experiment", "mnist", "--aggregator", "krum",
+              "--nb-workers", "5", "--nb-decl-byz-workers", "1",
+              "--nb-real-byz-workers", "1", "--attack", "reversal",
+              "--attack-args", "factor:2.0",
+              "--experiment-args", "batch-size:16",
+              "--max-step", "6", "--evaluation-delta", "-1",
+              "--evaluation-period", "-1", "--progress-every", "0"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+
+
+def test_runner_lossy_flags(tmp_path):
+    r = _run(["runner.py", "--experiment", "mnist", "--aggregator",
+              "average-nan", "--nb-workers", "4",
+              "--lossy", "drop-rate:0.2", "workers:0", "chunk-bytes:4096",
+              "--experiment-args", "batch-size:16",
+              "--max-step", "6", "--evaluation-delta", "-1",
+              "--evaluation-period", "-1", "--progress-every", "0"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+
+
+def test_runner_rejects_unknown_aggregator():
+    r = _run(["runner.py", "--experiment", "mnist", "--aggregator", "nope",
+              "--nb-workers", "2", "--max-step", "1"])
+    assert r.returncode != 0
+    assert b"Unknown GAR" in r.stderr or b"Unknown GAR" in r.stdout
+
+
+def test_deploy_local_two_ranks(tmp_path):
+    r = _run(["deploy.py", "--nproc", "2", "--master-port", "29701", "--",
+              "--experiment", "mnist", "--aggregator", "median",
+              "--nb-workers", "4", "--experiment-args", "batch-size:16",
+              "--max-step", "6", "--evaluation-delta", "-1",
+              "--evaluation-period", "-1", "--progress-every", "0"],
+             timeout=420)
+    assert r.returncode == 0, (r.stdout.decode()[-1500:]
+                               + r.stderr.decode()[-1500:])
+    assert "all ranks completed" in r.stdout.decode()
