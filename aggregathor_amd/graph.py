@@ -215,30 +215,36 @@ class Engine:
 
     def compute_local_gradients(self):
         """Fill ``local_rows`` with this rank's worker gradients; returns the
-        mean loss over the local workers."""
+        mean loss over the local workers (as a tensor)."""
+        import contextlib
         self.model.train()
         losses = []
-        for li, worker in enumerate(self.group.worker_ids):
-            row = self.local_rows[li]
-            row.zero_()
-            bind_grad_views(self.params, row)
-            batch = self.experiment.train_batch(worker, self.global_step, self.device)
-            self._trace(f"worker {worker}: forward")
-            if self.amp:
-                with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
-                    loss = self.experiment.loss(self.model, batch)
-            else:
+        # One autocast region for the whole worker loop: the bf16 weight
+        # casts are cached across the local workers' forwards instead of
+        # being re-materialized per micro-batch.
+        amp_ctx = (torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+                   if self.amp else contextlib.nullcontext())
+        with amp_ctx:
+            for li, worker in enumerate(self.group.worker_ids):
+                row = self.local_rows[li]
+                row.zero_()
+                bind_grad_views(self.params, row)
+                batch = self.experiment.train_batch(
+                    worker, self.global_step, self.device)
+                self._trace(f"worker {worker}: forward")
                 loss = self.experiment.loss(self.model, batch)
-            reg = self._regularization()
-            if reg is not None:
-                loss = loss + reg
-            self._trace(f"worker {worker}: backward")
-            loss.backward()
-            losses.append(loss.detach())
-            # Byzantine replacement: the worker computed honestly, then lies.
-            if self.attack is not None and worker < self.nb_real_byz:
-                self._trace(f"worker {worker}: byzantine craft")
-                row.copy_(self.attack.craft(row.clone(), worker, self.global_step))
+                if self.l1 > 0 or self.l2 > 0:
+                    with torch.autocast(device_type="cuda", enabled=False) \
+                            if self.amp else contextlib.nullcontext():
+                        loss = loss.float() + self._regularization()
+                self._trace(f"worker {worker}: backward")
+                loss.backward()
+                losses.append(loss.detach())
+                # Byzantine replacement: the worker computed honestly, then lies.
+                if self.attack is not None and worker < self.nb_real_byz:
+                    self._trace(f"worker {worker}: byzantine craft")
+                    row.copy_(self.attack.craft(row.clone(), worker,
+                                                self.global_step))
         return torch.stack(losses).mean()
 
     def aggregate(self):
@@ -262,13 +268,20 @@ class Engine:
         self.optimizer.step()
         self.global_step += 1
 
-    def step(self):
-        """One full training step; returns the local mean worker loss."""
+    def step(self, sync_loss=True):
+        """One full training step; returns the local mean worker loss.
+
+        ``sync_loss=False`` skips the device->host loss read (no implicit
+        synchronization -- benchmark hot loops use this; the NaN-divergence
+        check then only sees the value when one is requested).
+        """
         loss = self.compute_local_gradients()
         aggregated = self.aggregate()
         self.apply(aggregated)
-        self.last_loss = loss.item()
-        return self.last_loss
+        if sync_loss:
+            self.last_loss = loss.item()
+            return self.last_loss
+        return None
 
     # ------------------------------------------------------------------ #
 
