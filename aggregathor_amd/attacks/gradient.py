@@ -14,6 +14,8 @@ from .. import tools
 class ReversalAttack(_Attack):
     """Submit ``-factor x`` the honest gradient (gradient ascent)."""
 
+    graph_safe = True  # pure device math: hipGraph-capturable
+
     def __init__(self, args):
         self.args = tools.parse_keyval(args, defaults={"factor": 1.0})
 
@@ -37,6 +39,8 @@ class RandomAttack(_Attack):
 class ZeroAttack(_Attack):
     """Submit the zero gradient (stalling attack)."""
 
+    graph_safe = True
+
     def __init__(self, args):
         tools.parse_keyval(args)
 
@@ -48,6 +52,8 @@ class MagnitudeAttack(_Attack):
     """Submit an arbitrarily-scaled honest gradient (the classic unbounded
     attack Krum was designed against)."""
 
+    graph_safe = True
+
     def __init__(self, args):
         self.args = tools.parse_keyval(args, defaults={"factor": 1e6})
 
@@ -57,6 +63,8 @@ class MagnitudeAttack(_Attack):
 
 class NaNAttack(_Attack):
     """Submit all-NaN coordinates (tests non-finite handling end to end)."""
+
+    graph_safe = True
 
     def __init__(self, args):
         tools.parse_keyval(args)

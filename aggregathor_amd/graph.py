@@ -140,7 +140,8 @@ class Engine:
                  learning_rate="fixed", learning_rate_args=None,
                  l1_regularize=-1., l2_regularize=-1.,
                  nb_real_byz=0, attack=None, attack_args=None,
-                 lossy=None, amp=False, trace=False, seed=1234):
+                 lossy=None, amp=False, trace=False, seed=1234,
+                 use_graphs="auto", graph_warmup=3):
         """
         Args:
           experiment: an instantiated _Experiment
@@ -175,6 +176,7 @@ class Engine:
             aggregator, self.n, nbbyzwrks, aggregator_args or [])
         self.l1 = l1_regularize
         self.l2 = l2_regularize
+        self.lr_schedule_name = learning_rate
         self.lr_fn = build_lr_schedule(learning_rate, learning_rate_args or [])
         self.optimizer = build_optimizer(
             optimizer, optimizer_args or [], self.params, self.lr_fn(0))
@@ -196,6 +198,17 @@ class Engine:
         self.agg_flat = torch.zeros(self.d, dtype=torch.float32,
                                     device=self.device)
         self.last_loss = float("nan")
+
+        # hipGraph step capture (parallel/graphstep.py): auto-enabled on GPU
+        # for capturable configs, after `graph_warmup` eager steps.
+        from .parallel.graphstep import CapturedStep
+        self._graph_cls = CapturedStep
+        self.graph_warmup = graph_warmup
+        if use_graphs == "auto":
+            self.use_graphs = CapturedStep.supported(self)
+        else:
+            self.use_graphs = bool(use_graphs) and CapturedStep.supported(self)
+        self._graphstep = None
 
     # ------------------------------------------------------------------ #
 
@@ -268,6 +281,14 @@ class Engine:
         self.optimizer.step()
         self.global_step += 1
 
+    def _apply_from_matrix(self):
+        """Aggregate ``self.matrix`` and apply -- the capturable tail of a
+        step (no lr update, no step counter; the caller owns those)."""
+        aggregated = self.gar.aggregate(self.matrix)
+        self.agg_flat.copy_(aggregated)
+        bind_grad_views(self.params, self.agg_flat)
+        self.optimizer.step()
+
     def step(self, sync_loss=True):
         """One full training step; returns the local mean worker loss.
 
@@ -275,9 +296,21 @@ class Engine:
         synchronization -- benchmark hot loops use this; the NaN-divergence
         check then only sees the value when one is requested).
         """
-        loss = self.compute_local_gradients()
-        aggregated = self.aggregate()
-        self.apply(aggregated)
+        if self.use_graphs and self._graphstep is None \
+                and self.global_step >= self.graph_warmup:
+            self._trace("capturing hipGraph step")
+            # Freeze the (fixed) lr into the captured optimizer step.
+            lr = self.lr_fn(self.global_step)
+            for pg in self.optimizer.param_groups:
+                pg["lr"] = lr
+            self._graphstep = self._graph_cls(self)
+            self._graphstep.capture()
+        if self._graphstep is not None and self._graphstep.ready:
+            loss = self._graphstep.run()
+        else:
+            loss = self.compute_local_gradients()
+            aggregated = self.aggregate()
+            self.apply(aggregated)
         if sync_loss:
             self.last_loss = loss.item()
             return self.last_loss
@@ -304,3 +337,6 @@ class Engine:
         self.global_step = state["step"]
         self.model.load_state_dict(state["model"])
         self.optimizer.load_state_dict(state["optimizer"])
+        # A restore swaps optimizer-state storages: any captured graph holds
+        # stale pointers and must be rebuilt.
+        self._graphstep = None
