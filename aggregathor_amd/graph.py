@@ -298,15 +298,14 @@ class Engine:
         """
         if self.use_graphs and self._graphstep is None \
                 and self.global_step >= self.graph_warmup:
-            self._trace("capturing hipGraph step")
+            self._trace("engaging hipGraph step capture")
             # Freeze the (fixed) lr into the captured optimizer step.
             lr = self.lr_fn(self.global_step)
             for pg in self.optimizer.param_groups:
                 pg["lr"] = lr
             self._graphstep = self._graph_cls(self)
-            self._graphstep.capture()
-        if self._graphstep is not None and self._graphstep.ready:
-            loss = self._graphstep.run()
+        if self._graphstep is not None:
+            loss = self._graphstep.step_once()
         else:
             loss = self.compute_local_gradients()
             aggregated = self.aggregate()

@@ -32,11 +32,23 @@ def _attack_graph_safe(attack):
 
 
 class CapturedStep:
-    """Builds and replays the two-step hipGraph pair for an Engine."""
+    """Builds and replays the two-step hipGraph pair for an Engine.
+
+    State machine (one real training step per ``step_once`` call):
+      calls 1-2: real steps executed on the capture side stream (allocator
+                 warmup; trajectory identical to eager),
+      call 3:    record both graphs (recording executes nothing), then
+                 replay for this call's step,
+      later:     stage batches + replay.
+    """
+
+    WARM_STEPS = 2
 
     def __init__(self, engine):
         self.engine = engine
         self.ready = False
+        self.warmed = 0
+        self.static_batches = None
 
     @staticmethod
     def supported(engine):
@@ -45,24 +57,51 @@ class CapturedStep:
                 and _attack_graph_safe(engine.attack)
                 and engine.lr_schedule_name == "fixed")
 
-    def capture(self):
-        """Capture after the engine has run >= 1 eager warmup step."""
+    def _stage_batches(self):
+        """Copy this step's batches into the static input buffers (D2D)."""
         eng = self.engine
-        # Static input buffers (one per local worker).
-        self.static_batches = []
-        for worker in eng.group.worker_ids:
+        for li, worker in enumerate(eng.group.worker_ids):
             x, y = eng.experiment.train_batch(worker, eng.global_step,
                                               eng.device)
-            self.static_batches.append((x.clone(), y.clone()))
+            sx, sy = self.static_batches[li]
+            sx.copy_(x, non_blocking=True)
+            sy.copy_(y, non_blocking=True)
 
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for _ in range(2):  # warmup on the capture stream
-                self._local_phase()
+    def _ensure_buffers(self):
+        eng = self.engine
+        if self.static_batches is None:
+            self.static_batches = []
+            for worker in eng.group.worker_ids:
+                x, y = eng.experiment.train_batch(worker, eng.global_step,
+                                                  eng.device)
+                self.static_batches.append((x.clone(), y.clone()))
+            self.side = torch.cuda.Stream()
+
+    def step_once(self):
+        """One full training step; returns the loss tensor."""
+        eng = self.engine
+        self._ensure_buffers()
+        if self.warmed < self.WARM_STEPS:
+            # Real step on the capture stream: allocator warmup with the
+            # exact same training semantics as an eager step.
+            self._stage_batches()
+            self.side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self.side):
+                loss = self._local_phase()
+                if eng.group.distributed:
+                    eng.group.gather(eng.local_rows, out=eng.matrix)
                 eng._apply_from_matrix()
-        torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.current_stream().wait_stream(self.side)
+            eng.global_step += 1
+            self.warmed += 1
+            return loss
+        if not self.ready:
+            self._record()
+        return self.run()
 
+    def _record(self):
+        """Record both graphs (recording executes nothing)."""
+        eng = self.engine
         self.graph_local = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph_local):
             self.static_loss = self._local_phase()
@@ -100,12 +139,7 @@ class CapturedStep:
     def run(self):
         """One full training step via graph replay; returns the loss tensor."""
         eng = self.engine
-        for li, worker in enumerate(eng.group.worker_ids):
-            x, y = eng.experiment.train_batch(worker, eng.global_step,
-                                              eng.device)
-            sx, sy = self.static_batches[li]
-            sx.copy_(x, non_blocking=True)
-            sy.copy_(y, non_blocking=True)
+        self._stage_batches()
         self.graph_local.replay()
         if eng.group.distributed:
             eng.group.gather(eng.local_rows, out=eng.matrix)
