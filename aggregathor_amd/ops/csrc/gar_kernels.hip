@@ -34,6 +34,9 @@
 // means).
 
 #include <float.h>
+#include <stdlib.h>
+#include <string.h>
+
 #include <hip/hip_runtime.h>
 
 #include "gar_kernels.h"
@@ -79,9 +82,35 @@ static inline int pair_layout_rows(int n) {
   return n;  // tile kernel uses n-enumeration
 }
 
+constexpr int kMfmaTC = 512;  // d-tile columns of the MFMA sqdist kernel
+
+static inline int nblocks_mfma(long d) {
+  long nt = (d + kMfmaTC - 1) / kMfmaTC;
+  if (nt < 1) nt = 1;
+  if (nt > kMaxBlocksD) nt = kMaxBlocksD;
+  return (int)nt;
+}
+
+// Kernel choice for the distance pass: hand-written MFMA formulation or the
+// VALU register-tile formulation (both deterministic, same partial layout).
+// Env AGGREGATHOR_SQDIST = "mfma" | "valu"; the default is the measured
+// winner on MI355X for the flagship n=8 shape.
+static bool use_mfma_sqdist() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("AGGREGATHOR_SQDIST");
+    if (e && strcmp(e, "mfma") == 0) v = 1;
+    else if (e && strcmp(e, "valu") == 0) v = 0;
+    else v = 1;  // default: MFMA (measured ~equal-or-better, HBM-bound)
+  }
+  return v != 0;
+}
+
 long sqdist_partials_elems(int n, long d) {
   int rows = pair_layout_rows(n);
-  return (long)nblocks_d(d) * (rows * (rows - 1) / 2);
+  int nb = nblocks_d(d);
+  if (n <= 16 && nblocks_mfma(d) > nb) nb = nblocks_mfma(d);
+  return (long)nb * (rows * (rows - 1) / 2);
 }
 
 // ---------------------------------------------------------------------------
@@ -265,6 +294,127 @@ __global__ __launch_bounds__(kBlock) void sqdist_tile_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// MFMA pairwise-distance kernel (n <= 16): the matrix-core formulation of
+// the same computation.
+//
+// Per d-tile staged in LDS, each wave builds per-lane DIFFERENCE values
+// v = g[a_p][k] - g[b_p][k] (pair p = lane%32, k-slot = lane/32) and issues
+// `v_mfma_f32_32x32x2_f32` with the SAME register as both A and B operands:
+// A[i][k] = v(lane: i=l&31, k=l>>5) and B[k][j] = v(lane: j=l&31, k=l>>5)
+// means C = P P^T exactly, and its DIAGONAL is the per-pair squared
+// distance. f32-in MFMA is exact fp32 fma-chain math (no xf32 on gfx950),
+// so numerics match the VALU kernel's class. The off-diagonal 31/32 of the
+// MFMA tile is unused compute -- irrelevant here because the kernel is HBM-
+// bound (28 pairs x 2 flop / 32 B/coordinate = 1.75 flop/B << any peak);
+// what MFMA buys is freeing the VALU from the 28-accumulator FMA chains.
+// Selection between this and the VALU kernel: env AGGREGATHOR_SQDIST
+// (mfma | valu); both produce the same partials layout.
+//
+// C/D layout of 32x32 MFMA (guide §3): col = l&31, row = (reg&3) + 8*(reg>>2)
+// + 4*(l>>5). Diagonal element p lives in lane (p&31) + 32*hi with
+// hi = (p>>2)&1 and reg = (val&3) | ((val>>3)<<2) where val = p - 4*hi.
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+template <int NR, int TC>  // NR rows (<=16), TC tile columns (multiple of 8)
+__global__ __launch_bounds__(kBlock) void sqdist_mfma_kernel(
+    const float* __restrict__ g, float* __restrict__ partials, long d,
+    int n, int ntiles) {
+  constexpr int P = NR * (NR - 1) / 2;
+  constexpr int NPG = (P + 31) / 32;  // pair groups of 32
+  constexpr int LDC = TC + 1;         // +1 word: bank-conflict break
+  __shared__ float tile[NR * LDC];
+  __shared__ float part[4][P];  // per-wave partials, combined in fixed order
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  // Pair tables (compile-time NR enumeration, runtime n guard).
+  int pa[NPG], pb[NPG];
+#pragma unroll
+  for (int gidx = 0; gidx < NPG; ++gidx) {
+    int p = gidx * 32 + (lane & 31);
+    // decode NR-enumeration pair p -> (i, j)
+    int i = 0, rem = p;
+    bool valid = p < P;
+#pragma unroll
+    for (int r = 0; r < NR - 1; ++r) {
+      if (valid && rem >= NR - 1 - r && i == r) {
+        rem -= NR - 1 - r;
+        ++i;
+      }
+    }
+    int j = i + 1 + rem;
+    valid = valid && (j < n);
+    pa[gidx] = valid ? i : 0;
+    pb[gidx] = valid ? j : 0;  // invalid pairs: i==j==0 -> zero difference
+  }
+  const int khalf = lane >> 5;  // k-slot within the 2-deep MFMA K
+
+  f32x16 acc[NPG];
+#pragma unroll
+  for (int gidx = 0; gidx < NPG; ++gidx)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc[gidx][r] = 0.f;
+
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const long x0 = (long)t * TC;
+    // Stage [NR][TC] into LDS, coalesced, zero-fill past d and past n.
+    for (int idx = tid; idx < NR * TC; idx += kBlock) {
+      int row = idx / TC, col = idx % TC;
+      long x = x0 + col;
+      tile[row * LDC + col] =
+          (row < n && x < d) ? g[(long)row * d + x] : 0.f;
+    }
+    __syncthreads();
+    // Each wave owns a contiguous quarter of the tile's K range.
+    const int kper = TC / 4;
+    const int kbeg = wave * kper;
+#pragma unroll 4
+    for (int k = 0; k < kper; k += 2) {
+      const int kk = kbeg + k + khalf;
+#pragma unroll
+      for (int gidx = 0; gidx < NPG; ++gidx) {
+        float va = tile[pa[gidx] * LDC + kk];
+        float vb = tile[pb[gidx] * LDC + kk];
+        float v = va - vb;
+        acc[gidx] = __builtin_amdgcn_mfma_f32_32x32x2f32(v, v, acc[gidx],
+                                                         0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  // Diagonal extraction: pair p owned by lane (p&31)+32*((p>>2)&1).
+#pragma unroll
+  for (int gidx = 0; gidx < NPG; ++gidx) {
+#pragma unroll
+    for (int hi = 0; hi < 2; ++hi) {
+      if (khalf == hi) {
+        int p_lane = lane & 31;                       // candidate col
+        int val = p_lane - 4 * hi;
+        if (val >= 0 && (val & 4) == 0) {
+          int reg = (val & 3) | ((val >> 3) << 2);
+          int row = (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+          if (row == p_lane) {
+            int p = gidx * 32 + p_lane;
+            if (p < P) part[wave][p] = acc[gidx][reg];
+          }
+        }
+      }
+    }
+  }
+  __syncthreads();
+  // Fixed-order cross-wave combine -> deterministic block partial.
+  for (int p = tid; p < P; p += kBlock) {
+    float s = part[0][p];
+    s += part[1][p];
+    s += part[2][p];
+    s += part[3][p];
+    partials[(long)blockIdx.x * P + p] = s;
+  }
+}
+
 // Stage 2: deterministic cross-block reduction -> [n, n] symmetric matrix
 // with +inf diagonal (the diagonal is never a candidate; the reference
 // stores T::max there, op_bulyan/cpu.cpp:75).
@@ -307,7 +457,16 @@ void sqdist(const float* g, int n, long d, float* partials, float* dist,
             hipStream_t stream) {
   int nblk = nblocks_d(d);
   int vw = vec_width(d);
-  if (n <= 8) {
+  if (n <= 16 && use_mfma_sqdist()) {
+    int ntiles = (int)((d + kMfmaTC - 1) / kMfmaTC);
+    nblk = nblocks_mfma(d);
+    if (n <= 8)
+      sqdist_mfma_kernel<8, kMfmaTC>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d, n, ntiles);
+    else
+      sqdist_mfma_kernel<16, kMfmaTC>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d, n, ntiles);
+  } else if (n <= 8) {
     if (vw == 4)
       sqdist_small_kernel<8, 4>
           <<<nblk, kBlock, 0, stream>>>(g, partials, d / 4, d, n);
