@@ -101,7 +101,15 @@ static bool use_mfma_sqdist() {
     const char* e = getenv("AGGREGATHOR_SQDIST");
     if (e && strcmp(e, "mfma") == 0) v = 1;
     else if (e && strcmp(e, "valu") == 0) v = 0;
-    else v = 1;  // default: MFMA (measured ~equal-or-better, HBM-bound)
+    // Default: VALU. Measured on MI355X (n=8, d=25.6M fp32): VALU 0.200 ms
+    // (4.1 TB/s algorithmic) vs MFMA 0.606 ms. The f32-input MFMA runs at
+    // the f32 VECTOR rate (64 cyc/SIMD issue for 32x32x2, guide section 3)
+    // and only the 28-of-1024 diagonal outputs are useful, so the MFMA
+    // formulation is ISSUE-bound ~2.5x over HBM at small n, while the VALU
+    // kernel at 1.75 flop/B is purely HBM-bound. MFMA pairwise-L2 would pay
+    // only for bf16-stored gradients (2075 TF rate); the reference's GAR
+    // semantics are fp32.
+    else v = 0;
   }
   return v != 0;
 }
@@ -677,11 +685,30 @@ void bulyan_select(const float* dist, int n, int f, int m,
 // Per-coordinate rank-selection helpers (fully unrolled over NMAX so the
 // column stays in VGPRs; n is runtime <= NMAX).
 
+// Monotonic integer key for the total order: an ascending unsigned compare
+// of keys == lt_total on the values (non-finite -> max key, ties fall to the
+// index compare; -0.0 orders just before +0.0 which is observationally
+// equivalent since the values compare equal). Computing the key costs O(n)
+// per column ONCE, turning every rank comparison into 2 integer ops -- the
+// naive per-comparison isfinite chain made median/averaged-median VALU-bound
+// (measured 1.76 / 0.89 TB/s before, see profiles/).
+__device__ __forceinline__ unsigned sort_key(float v) {
+  if (!isfinite(v)) return 0xFFFFFFFFu;
+  unsigned u = __float_as_uint(v);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
 template <int NMAX>
 __device__ __forceinline__ float coord_median(const float (&vals)[NMAX],
                                               int n) {
   // Element at rank n/2 under the total order (native.cpp:686-694).
   const int target = n / 2;
+  unsigned keys[NMAX];
+#pragma unroll
+  for (int i = 0; i < NMAX; ++i) {
+    if (i >= n) break;
+    keys[i] = sort_key(vals[i]);
+  }
   float out = 0.f;
 #pragma unroll
   for (int i = 0; i < NMAX; ++i) {
@@ -690,7 +717,8 @@ __device__ __forceinline__ float coord_median(const float (&vals)[NMAX],
 #pragma unroll
     for (int j = 0; j < NMAX; ++j) {
       if (j >= n) break;
-      if (j != i && lt_total(vals[j], j, vals[i], i)) ++rank;
+      if (j != i)
+        rank += (keys[j] < keys[i]) | ((keys[j] == keys[i]) & (j < i));
     }
     if (rank == target) out = vals[i];
   }
@@ -702,11 +730,11 @@ __device__ __forceinline__ float coord_averaged_median(
     const float (&vals)[NMAX], int n, int beta) {
   // Mean of the beta elements closest to the median (native.cpp:714-739).
   const float zero = coord_median<NMAX>(vals, n);
-  float delta[NMAX];
+  unsigned dkeys[NMAX];
 #pragma unroll
   for (int i = 0; i < NMAX; ++i) {
     if (i >= n) break;
-    delta[i] = fabsf(vals[i] - zero);
+    dkeys[i] = sort_key(fabsf(vals[i] - zero));
   }
   float sum = 0.f;
 #pragma unroll
@@ -716,7 +744,8 @@ __device__ __forceinline__ float coord_averaged_median(
 #pragma unroll
     for (int j = 0; j < NMAX; ++j) {
       if (j >= n) break;
-      if (j != i && lt_total(delta[j], j, delta[i], i)) ++rank;
+      if (j != i)
+        rank += (dkeys[j] < dkeys[i]) | ((dkeys[j] == dkeys[i]) & (j < i));
     }
     if (rank < beta) sum += vals[i];
   }
