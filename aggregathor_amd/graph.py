@@ -141,7 +141,7 @@ class Engine:
                  l1_regularize=-1., l2_regularize=-1.,
                  nb_real_byz=0, attack=None, attack_args=None,
                  lossy=None, amp=False, trace=False, seed=1234,
-                 use_graphs="auto", graph_warmup=3):
+                 use_graphs="auto", graph_warmup=3, integrity=None):
         """
         Args:
           experiment: an instantiated _Experiment
@@ -161,6 +161,7 @@ class Engine:
         self.amp = amp and self.device.type == "cuda"
         self.trace = trace
         self.lossy = lossy
+        self.integrity = integrity
         self.global_step = 0
 
         torch.manual_seed(seed)
@@ -261,9 +262,20 @@ class Engine:
         return torch.stack(losses).mean()
 
     def aggregate(self):
-        """Gather all rows, inject channel loss, run the GAR."""
+        """Gather all rows, verify integrity, inject channel loss, run the
+        GAR."""
         self._trace("gather")
         matrix = self.group.gather(self.local_rows, out=self.matrix if self.group.distributed else None)
+        if self.integrity is not None:
+            self._trace("integrity check")
+            macs_local = self.integrity.sign_rows(
+                self.local_rows, self.group.worker_ids, self.global_step)
+            macs = torch.cat(self.group.gather_small(macs_local), dim=0)
+            failed = self.integrity.verify_matrix(matrix, macs,
+                                                  self.global_step)
+            if failed:
+                tools.warning(f"integrity check failed for workers {failed} "
+                              f"at step {self.global_step}; rows NaN-filled")
         if self.lossy is not None:
             self._trace("lossy inject")
             matrix = self.lossy.inject(matrix, self.global_step)

@@ -54,6 +54,7 @@ class CapturedStep:
     def supported(engine):
         return (engine.device.type == "cuda"
                 and engine.lossy is None
+                and engine.integrity is None  # host-side MAC computation
                 and _attack_graph_safe(engine.attack)
                 and engine.lr_schedule_name == "fixed")
 

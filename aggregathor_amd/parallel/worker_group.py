@@ -86,10 +86,21 @@ class WorkerGroup:
         for t in list(model.parameters()) + list(model.buffers()):
             dist.broadcast(t.data, src=0)
 
+    def gather_small(self, obj):
+        """All-gather a small per-rank object (e.g. MAC vectors); returns
+        the rank-ordered list."""
+        if self.world == 1:
+            return [obj]
+        objs = [None] * self.world
+        dist.all_gather_object(objs, obj)
+        return objs
+
     def allreduce_max(self, value):
         """Max over ranks of a Python float (used for worst-rank step time)."""
         if self.world == 1:
             return value
-        t = torch.tensor([value], dtype=torch.float64)
+        # RCCL reduces device tensors; gloo reduces host tensors.
+        dev = self.device if self.backend == "nccl" else "cpu"
+        t = torch.tensor([value], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         return t.item()

@@ -261,3 +261,4 @@ def can_access(path, read=True, write=False, recursive=False):
 
 
 from .checkpoint import Checkpoints  # noqa: E402,F401  (re-export)
+from .cluster import cluster_parse, cluster_parsers  # noqa: E402,F401

@@ -35,10 +35,10 @@ REPO = pathlib.Path(__file__).resolve().parent
 
 
 def parse_cluster(spec):
-    """Parse a JSON cluster spec: {"workers": ["host:port", ...]} (the
-    reference's format, tools/cluster.py:81-91; the ps/eval jobs have no
-    MI355X equivalent -- every rank is a worker)."""
-    data = json.loads(spec)
+    """Parse a cluster spec -- JSON {"workers": ["host:port", ...]} or a
+    special value like "G5k" (reference tools/cluster.py:81-91; the ps/eval
+    jobs have no MI355X equivalent -- every rank is a worker)."""
+    data = tools.cluster_parse(spec)
     hosts = []
     for job in ("workers", "ps", "local"):
         for entry in data.get(job, []):

@@ -70,6 +70,10 @@ parser.add_argument("--attack-args", nargs="*")
 parser.add_argument("--lossy", nargs="*", default=None,
                     help="UDP-style lossy-channel injection key:value args "
                          "(e.g. drop-rate:0.01 workers:0 clever:1)")
+parser.add_argument("--integrity-key", type=str, default="",
+                    help="Enable per-step gradient integrity MACs with this "
+                         "shared secret (the reference's message-signing "
+                         "equivalent; corrupted rows are NaN-filled)")
 parser.add_argument("--max-step", type=int, default=config.default_max_step)
 parser.add_argument("--checkpoint-dir", type=str, default="")
 parser.add_argument("--checkpoint-delta", type=int,
@@ -151,6 +155,10 @@ def main():
         lossy = None
         if args.lossy:
             lossy = LossyChannel(args.lossy)
+        integrity = None
+        if args.integrity_key:
+            from aggregathor_amd.parallel.signing import GradientIntegrity
+            integrity = GradientIntegrity(args.integrity_key, args.nb_workers)
         engine = Engine(
             experiment, args.aggregator, group,
             nbbyzwrks=args.nb_decl_byz_workers,
@@ -161,7 +169,7 @@ def main():
             l1_regularize=args.l1_regularize, l2_regularize=args.l2_regularize,
             nb_real_byz=args.nb_real_byz_workers, attack=args.attack,
             attack_args=args.attack_args or [], lossy=lossy, amp=args.amp,
-            trace=args.trace, seed=args.seed)
+            trace=args.trace, seed=args.seed, integrity=integrity)
         tools.info(f"model d = {engine.d} parameters, GAR = {args.aggregator}")
 
     with tools.Context("session", "info"):
