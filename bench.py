@@ -23,11 +23,17 @@ import os
 import sys
 import time
 
-# Avoid MIOpen exhaustive-search stalls on fresh boxes (kernels still cached
-# after the warmup steps, which are untimed).
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+# MIOpen configuration, measured on MI355X (profiles/gar_kernels.md):
+# FAST find mode falls back to workspace-starved col2im conv-backward
+# (~22 ms/step extra); normal find + torch benchmark mode + NHWC-friendly
+# kernels take the ResNet-50 n=8 step from 150 to 86 ms. The find cost is
+# paid once per conv config during the UNTIMED warmup steps.
+os.environ.pop("MIOPEN_FIND_MODE", None)
+os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 
 import torch
+
+torch.backends.cudnn.benchmark = True
 
 
 def main():

@@ -25,7 +25,8 @@ import os
 import pathlib
 import sys
 
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+# MIOpen conv-algo selection: see bench.py (same measured rationale).
+os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 
 from aggregathor_amd import config, tools
 
@@ -136,6 +137,8 @@ def main():
 
     with tools.Context("cluster", "info"):
         use_gpu = (args.use_gpu or args.reuse_gpu) and torch.cuda.is_available()
+        if use_gpu:
+            torch.backends.cudnn.benchmark = True
         if (args.use_gpu or args.reuse_gpu) and not torch.cuda.is_available():
             tools.warning("--use-gpu requested but no GPU is available; "
                           "falling back to CPU")
