@@ -12,7 +12,7 @@ configs are synthetic/random-init by definition).
 from . import _Experiment, register
 from .data import SyntheticClassification
 from .. import tools
-from ..models import RESNETS
+from ..models import NETWORKS
 
 _DATASETS = {
     "imagenet": {"shape": (3, 224, 224), "classes": 1000},
@@ -39,7 +39,7 @@ class ResNetExperiment(_Experiment):
             eval_examples=args["eval-examples"])
 
     def model(self):
-        return RESNETS[self.arch](num_classes=self.classes)
+        return NETWORKS[self.arch](num_classes=self.classes)
 
     def train_batch(self, worker, step, device):
         return self._synth.batch(self.args["batch-size"], worker, step, device)
@@ -56,11 +56,17 @@ def _make(arch, dataset):
     return _Bound
 
 
-for _arch in RESNETS:
+def _slim_alias(arch):
+    """Reference slim-name alias (experiments/slims.py naming convention)."""
+    if arch.startswith("resnet"):
+        return f"resnet_v1_{arch[len('resnet'):]}"
+    if arch.startswith("vgg"):
+        return f"vgg_{arch[len('vgg'):]}"
+    return arch
+
+
+for _arch in NETWORKS:
     for _ds in _DATASETS:
         cls = _make(_arch, _ds)
         register(f"{_arch}-{_ds}", cls)
-        # Reference slim-name alias, e.g. slim-resnet_v1_50-imagenet
-        # (experiments/slims.py naming convention).
-        _depth = _arch.replace("resnet", "")
-        register(f"slim-resnet_v1_{_depth}-{_ds}", cls)
+        register(f"slim-{_slim_alias(_arch)}-{_ds}", cls)

@@ -16,10 +16,20 @@ from .cnnet import CNNet
 from .resnet import (resnet18, resnet34, resnet50, resnet101, resnet152,
                      resnet200, cifar_resnet20, cifar_resnet32, cifar_resnet44,
                      cifar_resnet56, cifar_resnet110, RESNETS)
+from .vgg import vgg11, vgg16, vgg19
+from .mobilenet import mobilenet_v2
+
+#: Full model-factory map (the reference's slim ``networks_map`` analog,
+#: external/slim/nets/nets_factory.py:39-66).
+NETWORKS = dict(RESNETS)
+NETWORKS.update({
+    "vgg11": vgg11, "vgg16": vgg16, "vgg19": vgg19,
+    "mobilenet_v2": mobilenet_v2,
+})
 
 __all__ = [
-    "MLP", "CNNet", "RESNETS",
+    "MLP", "CNNet", "RESNETS", "NETWORKS",
     "resnet18", "resnet34", "resnet50", "resnet101", "resnet152", "resnet200",
     "cifar_resnet20", "cifar_resnet32", "cifar_resnet44", "cifar_resnet56",
-    "cifar_resnet110",
+    "cifar_resnet110", "vgg11", "vgg16", "vgg19", "mobilenet_v2",
 ]

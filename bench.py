@@ -120,7 +120,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,  # reference publishes no in-tree numbers
-            "dtype": "bf16" if amp else "fp32",
+            "dtype": "bf16" if engine.amp else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,

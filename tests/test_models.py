@@ -1,0 +1,66 @@
+"""Model zoo: shapes, experiment registry coverage."""
+
+import pytest
+import torch
+
+from aggregathor_amd import experiments
+from aggregathor_amd.models import (CNNet, MLP, NETWORKS, cifar_resnet20,
+                                    mobilenet_v2, resnet50, vgg11)
+
+
+def test_mlp_shape():
+    m = MLP()
+    assert m(torch.randn(4, 784)).shape == (4, 10)
+    assert m(torch.randn(4, 1, 28, 28)).shape == (4, 10)  # auto-flatten
+    assert sum(p.numel() for p in m.parameters()) == 784 * 100 + 100 + 100 * 10 + 10
+
+
+def test_cnnet_shape():
+    m = CNNet()
+    assert m(torch.randn(2, 3, 32, 32)).shape == (2, 10)
+
+
+@pytest.mark.parametrize("factory,shape,classes", [
+    (resnet50, (1, 3, 64, 64), 1000),
+    (cifar_resnet20, (2, 3, 32, 32), 10),
+    (vgg11, (1, 3, 64, 64), 1000),
+    (mobilenet_v2, (1, 3, 64, 64), 1000),
+])
+def test_network_forward(factory, shape, classes):
+    kw = {"num_classes": classes} if classes != 1000 else {}
+    m = factory(**kw)
+    out = m(torch.randn(*shape))
+    assert out.shape == (shape[0], classes)
+
+
+def test_resnet50_param_count():
+    # ~25.6M params, the flagship gradient dimension.
+    d = sum(p.numel() for p in resnet50().parameters())
+    assert 25_000_000 < d < 26_000_000
+
+
+def test_networks_map_complete():
+    for want in ("resnet18", "resnet50", "resnet101", "resnet152", "resnet200",
+                 "resnet20", "resnet110", "vgg11", "vgg16", "vgg19",
+                 "mobilenet_v2"):
+        assert want in NETWORKS
+
+
+def test_experiment_registry_coverage():
+    names = set(experiments.itemize())
+    for want in ("mnist", "mnistAttack", "cnnet",
+                 "resnet50-imagenet", "slim-resnet_v1_50-imagenet",
+                 "resnet20-cifar10", "resnet200-imagenet",
+                 "vgg16-imagenet", "slim-vgg_16-imagenet",
+                 "mobilenet_v2-imagenet", "slim-mobilenet_v2-imagenet"):
+        assert want in names, f"missing experiment {want!r}"
+
+
+def test_experiment_train_batch_device():
+    exp = experiments.instantiate("resnet20-cifar10",
+                                  ["batch-size:4", "eval-examples:8"])
+    x, y = exp.train_batch(0, 0, "cpu")
+    assert x.shape == (4, 3, 32, 32) and y.shape == (4,)
+    m = exp.model()
+    loss = exp.loss(m, (x, y))
+    assert loss.dim() == 0
