@@ -118,11 +118,18 @@ def flat_size(params):
 
 
 def bind_grad_views(params, flat):
-    """Bind each param's ``.grad`` as a view into the flat [d] buffer."""
+    """Bind each param's ``.grad`` as a view into the flat [d] buffer.
+
+    The view adopts the PARAM's stride layout (`as_strided`), so channels-
+    last conv weights get channels-last grad views: autograd accumulates
+    without a layout conversion, and the flat buffer simply stores that
+    param's gradient in NHWC element order -- identical on every rank, so
+    all GAR semantics are unchanged (a consistent coordinate permutation).
+    """
     off = 0
     for p in params:
         n = p.numel()
-        p.grad = flat[off:off + n].view_as(p)
+        p.grad = flat[off:off + n].as_strided(p.shape, p.stride())
         off += n
     assert off == flat.numel()
 
@@ -165,10 +172,14 @@ class Engine:
         self.global_step = 0
 
         torch.manual_seed(seed)
-        # Params stay contiguous: the flattened-gradient views bound before
-        # backward are contiguous, and a channels-last param would force a
-        # layout-converting grad accumulation on every parameter.
         self.model = experiment.model().to(self.device)
+        self.channels_last = (self.device.type == "cuda" and
+                              any(p.dim() == 4 for p in self.model.parameters()))
+        if self.channels_last:
+            # NHWC end to end: MIOpen's NHWC conv kernels skip the
+            # batched-transpose passes. Grad views adopt the param strides
+            # (bind_grad_views), so accumulation stays layout-matched.
+            self.model = self.model.to(memory_format=torch.channels_last)
         group.broadcast_model(self.model)
         self.params = [p for p in self.model.parameters() if p.requires_grad]
         self.d = flat_size(self.params)
@@ -217,6 +228,13 @@ class Engine:
         if self.trace:
             tools.trace(f"[step {self.global_step}] {msg}")
 
+    def _format_batch(self, batch):
+        if self.channels_last:
+            x, y = batch
+            if x.dim() == 4:
+                return x.to(memory_format=torch.channels_last), y
+        return batch
+
     def _regularization(self):
         """Reference graph.py:125-139: l1 = sum |w|; l2 = sqrt(sum w^2)."""
         reg = None
@@ -243,8 +261,8 @@ class Engine:
                 row = self.local_rows[li]
                 row.zero_()
                 bind_grad_views(self.params, row)
-                batch = self.experiment.train_batch(
-                    worker, self.global_step, self.device)
+                batch = self._format_batch(self.experiment.train_batch(
+                    worker, self.global_step, self.device))
                 self._trace(f"worker {worker}: forward")
                 loss = self.experiment.loss(self.model, batch)
                 if self.l1 > 0 or self.l2 > 0:

@@ -73,8 +73,8 @@ class CapturedStep:
         if self.static_batches is None:
             self.static_batches = []
             for worker in eng.group.worker_ids:
-                x, y = eng.experiment.train_batch(worker, eng.global_step,
-                                                  eng.device)
+                x, y = eng._format_batch(eng.experiment.train_batch(
+                    worker, eng.global_step, eng.device))
                 self.static_batches.append((x.clone(), y.clone()))
             self.side = torch.cuda.Stream()
 
