@@ -19,21 +19,25 @@ class SyntheticClassification:
     """Deterministic synthetic (inputs, labels) source."""
 
     def __init__(self, shape, classes, seed=1234, teacher=None,
-                 eval_examples=1024, scale=1.0):
+                 eval_examples=1024, scale=1.0, signal=0.5):
         """
         Args:
           shape:   per-example input shape, e.g. (784,) or (3, 32, 32)
           classes: number of classes
           seed:    base seed; all data derives deterministically from it
-          teacher: generate labels from a fixed random linear teacher when the
-                   flat dim is small (default: dim <= 8192); False = uniform
-                   random labels (throughput benchmarking)
+          teacher: True = labeled task (default when flat dim <= 8192):
+                   x = noise + signal * pattern[y] with one fixed random
+                   pattern per class -- a global-template detection task
+                   every architecture (MLPs AND pooling convnets) can learn.
+                   False = uniform random labels (throughput benchmarking).
+          signal:  class-pattern amplitude (task difficulty knob)
           eval_examples: size of the held-out deterministic eval set
         """
         self.shape = tuple(shape)
         self.classes = classes
         self.seed = seed
         self.scale = scale
+        self.signal = signal
         self.eval_examples = eval_examples
         dim = 1
         for s in self.shape:
@@ -42,9 +46,9 @@ class SyntheticClassification:
         self.teacher = (dim <= 8192) if teacher is None else teacher
         if self.teacher:
             gen = torch.Generator().manual_seed(seed ^ 0x7EAC4E12)
-            self.teacher_w = torch.randn(dim, classes, generator=gen)
+            self.patterns = torch.randn((classes, *self.shape), generator=gen)
         else:
-            self.teacher_w = None
+            self.patterns = None
         # Device-resident batch pool: on GPU, per-(worker, step) batches are
         # served from a once-generated pool of `pool_size` distinct batches
         # (batch(worker, step) == cpu batch(worker, step % pool_size)) --
@@ -53,17 +57,20 @@ class SyntheticClassification:
         self.pool_size = 8
         self._pools = {}
 
-    def _labels(self, flat_inputs, gen):
+    def _make(self, batch_size, gen):
+        """Generate (inputs, labels) from a generator."""
+        x = torch.randn((batch_size, *self.shape), generator=gen) * self.scale
         if self.teacher:
-            return (flat_inputs @ self.teacher_w).argmax(dim=1)
-        return torch.randint(0, self.classes, (flat_inputs.shape[0],), generator=gen)
+            y = torch.randint(0, self.classes, (batch_size,), generator=gen)
+            x = x + self.signal * self.patterns[y]
+        else:
+            y = torch.randint(0, self.classes, (batch_size,), generator=gen)
+        return x, y
 
     def _raw_batch(self, batch_size, worker, step):
         gen = torch.Generator().manual_seed(
             (self.seed * 1000003 + worker * 7919 + step * 104729) & 0x7FFFFFFF)
-        x = torch.randn((batch_size, *self.shape), generator=gen) * self.scale
-        y = self._labels(x.flatten(1), gen)
-        return x, y
+        return self._make(batch_size, gen)
 
     def batch(self, batch_size, worker, step, device="cpu"):
         """Training batch for (worker, step): pure function of the seed."""
@@ -87,7 +94,6 @@ class SyntheticClassification:
         remaining = self.eval_examples
         while remaining > 0:
             bs = min(batch_size, remaining)
-            x = torch.randn((bs, *self.shape), generator=gen) * self.scale
-            y = self._labels(x.flatten(1), gen)
+            x, y = self._make(bs, gen)
             yield x.to(device), y.to(device)
             remaining -= bs
