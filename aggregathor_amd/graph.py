@@ -193,6 +193,7 @@ class Engine:
         self.optimizer = build_optimizer(
             optimizer, optimizer_args or [], self.params, self.lr_fn(0))
 
+        self.phase_times = {}
         self.nb_real_byz = nb_real_byz
         self.attack = None
         if nb_real_byz > 0 and attack:
@@ -227,6 +228,17 @@ class Engine:
     def _trace(self, msg):
         if self.trace:
             tools.trace(f"[step {self.global_step}] {msg}")
+
+    def _phase(self, name, t0):
+        """Accumulate per-phase wall time (enabled with --trace; each phase
+        boundary synchronizes the device, so this perturbs throughput --
+        it is the reference's tf.Print phase bracketing analog)."""
+        import time
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        t1 = time.monotonic()
+        self.phase_times[name] = self.phase_times.get(name, 0.0) + (t1 - t0)
+        return t1
 
     def _format_batch(self, batch):
         if self.channels_last:
@@ -279,11 +291,12 @@ class Engine:
                                                 self.global_step))
         return torch.stack(losses).mean()
 
-    def aggregate(self):
-        """Gather all rows, verify integrity, inject channel loss, run the
-        GAR."""
-        self._trace("gather")
-        matrix = self.group.gather(self.local_rows, out=self.matrix if self.group.distributed else None)
+    def aggregate(self, matrix=None):
+        """Gather all rows (unless pre-gathered), verify integrity, inject
+        channel loss, run the GAR."""
+        if matrix is None:
+            self._trace("gather")
+            matrix = self.group.gather(self.local_rows, out=self.matrix if self.group.distributed else None)
         if self.integrity is not None:
             self._trace("integrity check")
             macs_local = self.integrity.sign_rows(
@@ -336,6 +349,19 @@ class Engine:
             self._graphstep = self._graph_cls(self)
         if self._graphstep is not None:
             loss = self._graphstep.step_once()
+        elif self.trace:
+            import time
+            t = time.monotonic()
+            loss = self.compute_local_gradients()
+            t = self._phase("local_gradients", t)
+            matrix = self.group.gather(
+                self.local_rows,
+                out=self.matrix if self.group.distributed else None)
+            t = self._phase("gather", t)
+            aggregated = self.aggregate(matrix)
+            t = self._phase("aggregate", t)
+            self.apply(aggregated)
+            self._phase("apply", t)
         else:
             loss = self.compute_local_gradients()
             aggregated = self.aggregate()

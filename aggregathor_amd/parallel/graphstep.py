@@ -55,6 +55,7 @@ class CapturedStep:
         return (engine.device.type == "cuda"
                 and engine.lossy is None
                 and engine.integrity is None  # host-side MAC computation
+                and not engine.trace          # --trace wants phase timing
                 and _attack_graph_safe(engine.attack)
                 and engine.lr_schedule_name == "fixed")
 

@@ -207,4 +207,9 @@ class Trainer:
                 tools.info(f"Off-step time: {report['off_step_time_s']:.3f} s")
                 tools.info(f"Steps/s (all steps): {report['steps_per_sec_all']:.3f}")
                 tools.info(f"Steps/s (excluding first): {report['steps_per_sec_excl_first']:.3f}")
+                if getattr(engine, "phase_times", None):
+                    for name, secs in engine.phase_times.items():
+                        tools.info(f"Phase {name}: {secs:.3f} s "
+                                   f"({1e3 * secs / max(steps_done, 1):.2f} ms/step)")
+        report["phase_times"] = dict(getattr(engine, "phase_times", {}))
         return report

@@ -178,3 +178,12 @@ def test_l2_regularization_changes_gradients():
     l1 = e1.step()
     l2 = e2.step()
     assert l2 > l1  # reg adds a positive term
+
+
+def test_trace_phase_times(capsys):
+    eng = _make_engine(trace=True)
+    for _ in range(3):
+        eng.step()
+    assert set(eng.phase_times) == {"local_gradients", "gather", "aggregate",
+                                    "apply"}
+    assert all(v > 0 for v in eng.phase_times.values())
