@@ -13,6 +13,8 @@ under bf16 autocast with fp32 master weights and fp32 flattened gradients
 
 import torch.nn as nn
 
+from .norm import norm2d
+
 
 def _conv3(cin, cout, stride=1):
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
@@ -28,9 +30,9 @@ class BasicBlock(nn.Module):
     def __init__(self, cin, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = _conv3(cin, planes, stride)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = norm2d(planes)
         self.conv2 = _conv3(planes, planes)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = norm2d(planes)
         self.act = nn.ReLU(inplace=True)
         self.downsample = downsample
 
@@ -47,11 +49,11 @@ class Bottleneck(nn.Module):
     def __init__(self, cin, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = _conv1(cin, planes)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = norm2d(planes)
         self.conv2 = _conv3(planes, planes, stride)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = norm2d(planes)
         self.conv3 = _conv1(planes, planes * 4)
-        self.bn3 = nn.BatchNorm2d(planes * 4)
+        self.bn3 = norm2d(planes * 4)
         self.act = nn.ReLU(inplace=True)
         self.downsample = downsample
 
@@ -70,7 +72,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.inplanes = 64
         self.conv1 = nn.Conv2d(in_ch, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = norm2d(64)
         self.act = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._stage(block, 64, layers[0])
@@ -86,7 +88,7 @@ class ResNet(nn.Module):
         if stride != 1 or self.inplanes != planes * block.expansion:
             downsample = nn.Sequential(
                 _conv1(self.inplanes, planes * block.expansion, stride),
-                nn.BatchNorm2d(planes * block.expansion))
+                norm2d(planes * block.expansion))
         stage = [block(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
@@ -117,7 +119,7 @@ class CifarResNet(nn.Module):
         n = (depth - 2) // 6
         self.inplanes = 16
         self.conv1 = _conv3(in_ch, 16)
-        self.bn1 = nn.BatchNorm2d(16)
+        self.bn1 = norm2d(16)
         self.act = nn.ReLU(inplace=True)
         self.layer1 = self._stage(16, n)
         self.layer2 = self._stage(32, n, stride=2)
@@ -135,7 +137,7 @@ class CifarResNet(nn.Module):
         downsample = None
         if stride != 1 or self.inplanes != planes:
             downsample = nn.Sequential(
-                _conv1(self.inplanes, planes, stride), nn.BatchNorm2d(planes))
+                _conv1(self.inplanes, planes, stride), norm2d(planes))
         stage = [BasicBlock(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes
         for _ in range(1, blocks):

@@ -88,7 +88,7 @@ def main():
     engine = Engine(
         exp, args.gar, group, nbbyzwrks=args.f, amp=amp,
         nb_real_byz=(args.f if args.attack else 0), attack=args.attack,
-        optimizer="sgd", learning_rate="fixed")
+        optimizer="sgd", learning_rate="fixed", graph_warmup=1)
 
     def sync():
         if device.startswith("cuda"):
@@ -97,6 +97,14 @@ def main():
 
     for _ in range(args.warmup):
         engine.step()
+    # hipGraph capture must never land in the timed region: run extra
+    # (untimed) steps until the capture has engaged, whatever warmup the
+    # driver chose.
+    extra = 0
+    while engine.use_graphs and (engine._graphstep is None
+                                 or not engine._graphstep.ready) and extra < 8:
+        engine.step()
+        extra += 1
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
