@@ -348,7 +348,19 @@ class Engine:
                 pg["lr"] = lr
             self._graphstep = self._graph_cls(self)
         if self._graphstep is not None:
-            loss = self._graphstep.step_once()
+            try:
+                loss = self._graphstep.step_once()
+            except RuntimeError as e:
+                # Capture can fail on exotic configs; training must not.
+                # Fall back to eager permanently (the warm steps it already
+                # ran were full training steps, so the trajectory is intact).
+                tools.warning(f"hipGraph step failed ({e}); falling back to "
+                              f"eager execution")
+                self.use_graphs = False
+                self._graphstep = None
+                loss = self.compute_local_gradients()
+                aggregated = self.aggregate()
+                self.apply(aggregated)
         elif self.trace:
             import time
             t = time.monotonic()

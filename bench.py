@@ -50,6 +50,8 @@ def main():
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--attack", type=str, default="",
                     help="optional attack (e.g. reversal) mounted by f workers")
+    ap.add_argument("--lossy", nargs="*", default=None,
+                    help="UDP-style lossy-channel injection key:value args")
     ap.add_argument("--device", type=str, default="",
                     help="override device (debug; cpu allowed)")
     ap.add_argument("--no-amp", action="store_true",
@@ -85,10 +87,14 @@ def main():
          "eval-examples:0"])
     group = WorkerGroup(args.workers, device=device)
     amp = not args.no_amp
+    lossy = None
+    if args.lossy:
+        from aggregathor_amd.attacks.lossy import LossyChannel
+        lossy = LossyChannel(args.lossy)
     engine = Engine(
         exp, args.gar, group, nbbyzwrks=args.f, amp=amp,
         nb_real_byz=(args.f if args.attack else 0), attack=args.attack,
-        optimizer="sgd", learning_rate="fixed", graph_warmup=1)
+        optimizer="sgd", learning_rate="fixed", graph_warmup=1, lossy=lossy)
 
     def sync():
         if device.startswith("cuda"):
@@ -112,6 +118,12 @@ def main():
     sync()
     elapsed = time.perf_counter() - t0
     elapsed = group.allreduce_max(elapsed)  # worst rank defines job time
+    # Validity: one synchronous loss read after timing -- a NaN here means
+    # the measured steps were doing garbage math.
+    final_loss = engine.step()
+    if final_loss != final_loss:
+        print(f"[bench] WARNING: post-run loss is NaN (rank {rank})",
+              file=sys.stderr)
 
     steps_per_sec = args.steps / elapsed
     global_batch = args.batch_size * args.workers
@@ -140,6 +152,7 @@ def main():
                 "f": args.f,
                 "attack": args.attack or None,
                 "images_per_sec": steps_per_sec * global_batch,
+                "final_loss": final_loss,
             },
         }
         print(json.dumps(result), flush=True)
