@@ -42,6 +42,8 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--model", type=str, default="resnet50")
+    ap.add_argument("--dataset", type=str, default="imagenet",
+                    help="dataset shape: imagenet | cifar10")
     ap.add_argument("--workers", type=int, default=8, help="total GAR workers n")
     ap.add_argument("--f", type=int, default=2, help="declared Byzantine workers")
     ap.add_argument("--gar", type=str, default="krum",
@@ -81,9 +83,10 @@ def main():
         raise RuntimeError("HIP extension _gar_hip not built -- run "
                            "`python -m aggregathor_amd.ops.build` first")
 
+    image_size = args.image_size if args.dataset == "imagenet" else 0
     exp = experiments.instantiate(
-        f"{args.model}-imagenet",
-        [f"batch-size:{args.batch_size}", f"image-size:{args.image_size}",
+        f"{args.model}-{args.dataset}",
+        [f"batch-size:{args.batch_size}", f"image-size:{image_size}",
          "eval-examples:0"])
     group = WorkerGroup(args.workers, device=device)
     amp = not args.no_amp
