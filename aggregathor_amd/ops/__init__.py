@@ -46,6 +46,13 @@ def hip_available():
     return _load_extension() is not None
 
 
+def itemize():
+    """Names of the available GAR compute ops (the reference's
+    ``native.itemize_op`` analog, native/__init__.py:374-377)."""
+    return ["pairwise_sqdist", "krum", "bulyan", "median", "averaged_median",
+            "average_nan", "average", "selection_average"]
+
+
 def _want_hip(tensor):
     if tensor.device.type != "cuda":
         return False

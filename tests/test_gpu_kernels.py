@@ -181,3 +181,39 @@ def test_determinism_bitwise():
     a = ext.bulyan(_rand(11, 1 << 18, seed=34), 2, 8)
     b = ext.bulyan(_rand(11, 1 << 18, seed=34), 2, 8)
     assert torch.equal(a, b)
+
+
+# ---------------------------------------------------------------------------- #
+# Edge shapes on device
+
+
+@pytest.mark.parametrize("n,d", [(4, 1), (5, 3), (8, 7), (8, 255),
+                                 (2, 1024)])
+def test_gpu_tiny_shapes(n, d):
+    ext = _require_ext()
+    g = _rand(n, d, seed=n * 31 + d)
+    got = ext.pairwise_sqdist(g).cpu()
+    want = _ref64(R.pairwise_sqdist, g.cpu())
+    off = ~torch.eye(n, dtype=torch.bool)
+    torch.testing.assert_close(got[off], want[off], rtol=1e-5, atol=1e-5)
+    got_m = ext.median(g).cpu()
+    assert torch.equal(got_m, R.median(g.cpu()))
+
+
+@pytest.mark.parametrize("n,f", [(40, 9), (64, 10)])
+def test_gpu_large_n_krum(n, f):
+    ext = _require_ext()
+    g = _rand(n, 30000, seed=n)
+    m = n - f - 2
+    got = ext.krum(g, f, m).cpu()
+    want = _ref64(R.krum, g.cpu(), f, m)
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-6)
+
+
+def test_gpu_identical_rows():
+    ext = _require_ext()
+    g = torch.ones((8, 100000), device="cuda")
+    out = ext.krum(g, 2, 4).cpu()
+    torch.testing.assert_close(out, torch.ones(100000), rtol=1e-6, atol=0)
+    out = ext.bulyan(torch.ones((11, 50000), device="cuda"), 2, 7).cpu()
+    torch.testing.assert_close(out, torch.ones(50000), rtol=1e-6, atol=0)
