@@ -21,8 +21,6 @@ set (adadelta / adagrad / adam / rmsprop / sgd) mirror graph.py:51-66 with
 the same sub-argument names and defaults.
 """
 
-import math
-
 import torch
 
 from . import aggregators as aggregators_mod

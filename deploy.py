@@ -21,7 +21,6 @@ Examples:
 """
 
 import argparse
-import json
 import os
 import pathlib
 import shlex

@@ -22,7 +22,6 @@ Flag mapping notes:
 
 import argparse
 import os
-import pathlib
 import sys
 
 # MIOpen conv-algo selection: see bench.py (same measured rationale).

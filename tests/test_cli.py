@@ -7,8 +7,6 @@ import pathlib
 import subprocess
 import sys
 
-import pytest
-
 REPO = pathlib.Path(__file__).resolve().parent.parent
 
 

@@ -1,7 +1,6 @@
 """HIP kernel numerics on a real MI355X: every GAR kernel vs the fp64-cast
 PyTorch oracle, including NaN cases, plus the fail-loud native-path check."""
 
-import numpy as np
 import pytest
 import torch
 
