@@ -192,6 +192,7 @@ class Engine:
             optimizer, optimizer_args or [], self.params, self.lr_fn(0))
 
         self.phase_times = {}
+        self._eval_model = None
         self.nb_real_byz = nb_real_byz
         self.attack = None
         if nb_real_byz > 0 and attack:
@@ -384,12 +385,19 @@ class Engine:
     # ------------------------------------------------------------------ #
 
     def evaluate(self):
-        """Top-1 accuracy on the experiment's eval set (rank-0 buffers)."""
-        if self.group.distributed:
-            import torch.distributed as dist
-            for b in self.model.buffers():
-                dist.broadcast(b.data, src=0)
-        return self.experiment.accuracy(self.model, self.device)
+        """Top-1 accuracy on the experiment's eval set.
+
+        Runs on a cached REPLICA of the model: the evaluation service thread
+        must neither flip the shared model's train/eval mode while the
+        training thread is mid-forward, nor issue collectives (it runs on
+        rank 0 only -- a broadcast here would deadlock the other ranks).
+        BN running stats are the local rank's, exactly like the reference's
+        eval replicas reading the PS variables concurrently with updates.
+        """
+        if self._eval_model is None:
+            self._eval_model = self.experiment.model().to(self.device)
+        self._eval_model.load_state_dict(self.model.state_dict())
+        return self.experiment.accuracy(self._eval_model, self.device)
 
     def state_dict(self):
         return {
