@@ -19,7 +19,8 @@ OPS_DIR = pathlib.Path(__file__).resolve().parent
 CSRC = OPS_DIR / "csrc"
 OUT = OPS_DIR / "_gar_hip.so"
 
-SOURCES = [CSRC / "gar_ops.cpp", CSRC / "gar_kernels.hip"]
+SOURCES = [CSRC / "gar_ops.cpp", CSRC / "gar_kernels.hip",
+           CSRC / "bn_kernels.hip"]
 HEADERS = [CSRC / "gar_kernels.h"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")

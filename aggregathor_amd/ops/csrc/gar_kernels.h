@@ -53,4 +53,22 @@ void averaged_median(const float* g, int n, long d, int beta, float* out,
                      hipStream_t stream);
 void average_nan(const float* g, int n, long d, float* out, hipStream_t stream);
 
+// ---------------------------------------------------------------------------
+// Fused NHWC BatchNorm training kernels (bn_kernels.hip). Data is a flat
+// [M, C] channels-last matrix; dtype tag 0 = fp32, 1 = bf16; stats fp32.
+
+// fp32 elements of the per-block partials workspace for C channels.
+long bn_partials_elems(int c);
+
+void bn_fwd(const void* x, void* y, long m, int c, int dtype, float eps,
+            float momentum, const float* weight, const float* bias,
+            float* running_mean, float* running_var, float* mean,
+            float* invstd, float* partials, hipStream_t stream);
+
+// consts: 3*C fp32 scratch for the per-channel dx constants.
+void bn_bwd(const void* dy, const void* x, void* dx, long m, int c,
+            int dtype, const float* weight, const float* mean,
+            const float* invstd, float* dweight, float* dbias,
+            float* partials, float* consts, hipStream_t stream);
+
 }  // namespace gar

@@ -129,7 +129,74 @@ torch::Tensor selection_average(torch::Tensor g, torch::Tensor sel) {
   return out;
 }
 
+// ---------------------------------------------------------------------------
+// Fused NHWC BatchNorm (training). x: [N, C, H, W] channels_last (bf16 or
+// fp32); weight/bias/running stats fp32. Returns (y, save_mean,
+// save_invstd).
+
+static void check_bn_input(const torch::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn: expected 4D GPU tensor");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn: expected channels_last input");
+  TORCH_CHECK(x.size(1) % 4 == 0, "bn: C must be a multiple of 4");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32 ||
+                  x.scalar_type() == torch::kBFloat16,
+              "bn: fp32 or bf16 input");
+}
+
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
+                                        torch::Tensor bias,
+                                        torch::Tensor running_mean,
+                                        torch::Tensor running_var,
+                                        double momentum, double eps) {
+  check_bn_input(x);
+  const long m = x.size(0) * x.size(2) * x.size(3);
+  const int c = (int)x.size(1);
+  const int dtype = x.scalar_type() == torch::kBFloat16 ? 1 : 0;
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({c}, f32);
+  auto invstd = torch::empty({c}, f32);
+  auto partials = torch::empty({gar::bn_partials_elems(c)}, f32);
+  gar::bn_fwd(x.data_ptr(), y.data_ptr(), m, c, dtype, (float)eps,
+              (float)momentum, weight.data_ptr<float>(),
+              bias.data_ptr<float>(), running_mean.data_ptr<float>(),
+              running_var.data_ptr<float>(), mean.data_ptr<float>(),
+              invstd.data_ptr<float>(), partials.data_ptr<float>(),
+              current_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_bwd_train(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor weight,
+                                        torch::Tensor mean,
+                                        torch::Tensor invstd) {
+  check_bn_input(x);
+  TORCH_CHECK(dy.sizes() == x.sizes() && dy.scalar_type() == x.scalar_type(),
+              "bn_bwd: dy/x mismatch");
+  auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  const long m = x.size(0) * x.size(2) * x.size(3);
+  const int c = (int)x.size(1);
+  const int dtype = x.scalar_type() == torch::kBFloat16 ? 1 : 0;
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto dx = torch::empty_like(x);
+  auto dweight = torch::empty({c}, f32);
+  auto dbias = torch::empty({c}, f32);
+  auto partials = torch::empty({gar::bn_partials_elems(c)}, f32);
+  auto consts = torch::empty({3L * c}, f32);
+  gar::bn_bwd(dyc.data_ptr(), x.data_ptr(), dx.data_ptr(), m, c, dtype,
+              weight.data_ptr<float>(), mean.data_ptr<float>(),
+              invstd.data_ptr<float>(), dweight.data_ptr<float>(),
+              dbias.data_ptr<float>(), partials.data_ptr<float>(),
+              consts.data_ptr<float>(), current_stream());
+  return {dx, dweight, dbias};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("bn_fwd_train", &bn_fwd_train,
+          "Fused NHWC BatchNorm training forward (gfx950)");
+  mod.def("bn_bwd_train", &bn_bwd_train,
+          "Fused NHWC BatchNorm training backward (gfx950)");
   mod.def("pairwise_sqdist", &pairwise_sqdist,
           "All-pairs squared L2 distances [n,n] (diag=+inf), gfx950 kernels");
   mod.def("krum", &krum, "Multi-Krum GAR (gfx950 kernels)");
