@@ -182,12 +182,16 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_norm_kernel(
   }
   __syncthreads();
   const int nq = c / 4;
-  const long total = m * nq;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    long row = i / nq;
-    int qt = (int)(i - row * nq);
+  // Incremental (row, quad) tracking: one 64-bit division per thread
+  // instead of one per element (64-bit div is emulated and dominated the
+  // first version of this kernel).
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long row = i0 / nq;
+  int qt = (int)(i0 - row * nq);
+  const long dr = stride / nq;
+  const int dq = (int)(stride - dr * nq);
+  while (row < m) {
     tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
     tvec4<T> o;
 #pragma unroll
@@ -196,6 +200,12 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_norm_kernel(
       o.v[k] = from_f32<T>(fmaf(to_f32<T>(v.v[k]), sa[ch], sb[ch]));
     }
     reinterpret_cast<tvec4<T>*>(y + row * c)[qt] = o;
+    qt += dq;
+    row += dr;
+    if (qt >= nq) {
+      qt -= nq;
+      row += 1;
+    }
   }
 }
 
@@ -334,12 +344,13 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_dx_kernel(
   }
   __syncthreads();
   const int nq = c / 4;
-  const long total = m * nq;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    long row = i / nq;
-    int qt = (int)(i - row * nq);
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long row = i0 / nq;
+  int qt = (int)(i0 - row * nq);
+  const long dr = stride / nq;
+  const int dq = (int)(stride - dr * nq);
+  while (row < m) {
     tvec4<T> g = reinterpret_cast<const tvec4<T>*>(dy + row * c)[qt];
     tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
     tvec4<T> o;
@@ -351,6 +362,12 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_dx_kernel(
       o.v[k] = from_f32<T>(r);
     }
     reinterpret_cast<tvec4<T>*>(dx + row * c)[qt] = o;
+    qt += dq;
+    row += dr;
+    if (qt >= nq) {
+      qt -= nq;
+      row += 1;
+    }
   }
 }
 
