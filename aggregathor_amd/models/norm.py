@@ -43,7 +43,15 @@ class _FusedBNFunction(torch.autograd.Function):
 class FusedBatchNorm2d(nn.BatchNorm2d):
     """BatchNorm2d using the fused NHWC HIP kernels when applicable
     (training, GPU, channels_last, C % 4 == 0, bf16/fp32); falls back to
-    the stock implementation otherwise (eval mode, CPU, odd shapes)."""
+    the stock implementation otherwise (eval mode, CPU, odd shapes).
+
+    EXPERIMENTAL (AGGREGATHOR_BN=fused, default off): the forward computes
+    variance single-pass as E[x^2] - E[x]^2, which can cancel
+    catastrophically for channels with |mean| >> sigma at large M (observed
+    as late-step divergence on ResNet-50 @ 224; fine at <= 112). v2 needs a
+    Welford/two-pass stats kernel. MIOpen's BN also measured faster per
+    layer (profiles/NOTES.md), so this path is a numerics-tested starting
+    point for the fusion work, not the default."""
 
     def forward(self, x):
         if (self.training and x.is_cuda and x.dim() == 4
@@ -73,7 +81,7 @@ class NativeBatchNorm2d(nn.BatchNorm2d):
 def norm2d(channels):
     """BatchNorm2d factory honoring AGGREGATHOR_BN (fused | native | miopen)."""
     kind = os.environ.get("AGGREGATHOR_BN", "miopen")
-    if kind == "fused":
+    if kind == "fused":  # EXPERIMENTAL -- see FusedBatchNorm2d docstring
         return FusedBatchNorm2d(channels)
     if kind == "native":
         return NativeBatchNorm2d(channels)
