@@ -45,13 +45,12 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
     (training, GPU, channels_last, C % 4 == 0, bf16/fp32); falls back to
     the stock implementation otherwise (eval mode, CPU, odd shapes).
 
-    EXPERIMENTAL (AGGREGATHOR_BN=fused, default off): the forward computes
-    variance single-pass as E[x^2] - E[x]^2, which can cancel
-    catastrophically for channels with |mean| >> sigma at large M (observed
-    as late-step divergence on ResNet-50 @ 224; fine at <= 112). v2 needs a
-    Welford/two-pass stats kernel. MIOpen's BN also measured faster per
-    layer (profiles/NOTES.md), so this path is a numerics-tested starting
-    point for the fusion work, not the default."""
+    EXPERIMENTAL (AGGREGATHOR_BN=fused, default off): variance is computed
+    single-pass SHIFTED by the running mean (exact where mean drift
+    develops; the naive E[x^2]-E[x]^2 form diverged on ResNet-50 @ 224).
+    MIOpen's BN measured faster per layer (profiles/NOTES.md), so this path
+    is a numerics-tested starting point for fusion work, not the
+    default."""
 
     def forward(self, x):
         if (self.training and x.is_cuda and x.dim() == 4

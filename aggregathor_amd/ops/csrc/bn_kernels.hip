@@ -70,9 +70,15 @@ static inline int bn_nblocks(long total_quads) {
 // of sub-threads in a fixed order -> deterministic. Loads are tvec4
 // (coalesced along C).
 
+// Variance is accumulated SHIFTED by the channel's running mean
+// (sum (x - s), sum (x - s)^2): the naive E[x^2] - E[x]^2 form cancels
+// catastrophically once a channel's |mean| >> sigma; the running mean
+// tracks the batch mean after a few steps, so the shifted form stays
+// exact precisely when drift develops. shift == nullptr -> 0.
 template <typename T>
 __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
-    const T* __restrict__ x, float* __restrict__ partials, long m, int c) {
+    const T* __restrict__ x, float* __restrict__ partials, long m, int c,
+    const float* __restrict__ shift) {
   const int nq = c / 4;
   __shared__ float red[2][kBnBlock * 4];
 
@@ -81,13 +87,17 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
     for (int q0 = 0; q0 < nq; q0 += kBnBlock) {
       int qt = q0 + (int)threadIdx.x;
       if (qt >= nq) break;
+      float sh[4];
+#pragma unroll
+      for (int k = 0; k < 4; ++k)
+        sh[k] = shift ? shift[qt * 4 + k] : 0.f;
       float s[4] = {0.f, 0.f, 0.f, 0.f};
       float q2[4] = {0.f, 0.f, 0.f, 0.f};
       for (long row = blockIdx.x; row < m; row += gridDim.x) {
         tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
 #pragma unroll
         for (int k = 0; k < 4; ++k) {
-          float f = to_f32<T>(v.v[k]);
+          float f = to_f32<T>(v.v[k]) - sh[k];
           s[k] += f;
           q2[k] = fmaf(f, f, q2[k]);
         }
@@ -105,6 +115,9 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
   const int qt = (int)threadIdx.x % nq;
   const int sub = (int)threadIdx.x / nq;
   const int nsub = kBnBlock / nq;
+  float sh[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k) sh[k] = shift ? shift[qt * 4 + k] : 0.f;
   float s[4] = {0.f, 0.f, 0.f, 0.f};
   float q2[4] = {0.f, 0.f, 0.f, 0.f};
   if (sub < nsub) {
@@ -114,7 +127,7 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
       tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
 #pragma unroll
       for (int k = 0; k < 4; ++k) {
-        float f = to_f32<T>(v.v[k]);
+        float f = to_f32<T>(v.v[k]) - sh[k];
         s[k] += f;
         q2[k] = fmaf(f, f, q2[k]);
       }
@@ -147,7 +160,8 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
 __global__ void bn_fwd_finalize_kernel(
     const float* __restrict__ partials, int nblk, int c, long m, float eps,
     float momentum, float* __restrict__ mean, float* __restrict__ invstd,
-    float* __restrict__ running_mean, float* __restrict__ running_var) {
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    const float* __restrict__ shift) {
   int ch = blockIdx.x * blockDim.x + threadIdx.x;
   if (ch >= c) return;
   float s = 0.f, q = 0.f;
@@ -155,8 +169,9 @@ __global__ void bn_fwd_finalize_kernel(
     s += partials[((long)b * 2) * c + ch];
     q += partials[((long)b * 2 + 1) * c + ch];
   }
-  float mu = s / (float)m;
-  float var = fmaxf(q / (float)m - mu * mu, 0.f);
+  float d = s / (float)m;                     // mean of (x - shift)
+  float var = fmaxf(q / (float)m - d * d, 0.f);
+  float mu = d + (shift ? shift[ch] : 0.f);
   mean[ch] = mu;
   invstd[ch] = rsqrtf(var + eps);
   if (running_mean != nullptr) {
@@ -388,11 +403,13 @@ static void bn_fwd_t(const T* x, T* y, long m, int c, float eps,
   const int nq = c / 4;
   const int nsub = kBnBlock / nq > 0 ? kBnBlock / nq : 1;
   int nblk = bn_nblocks((m + nsub - 1) / nsub);
+  // The running mean doubles as the per-channel variance shift.
+  const float* shift = running_mean;
   bn_fwd_partial_kernel<T>
-      <<<nblk, kBnBlock, 0, stream>>>(x, partials, m, c);
+      <<<nblk, kBnBlock, 0, stream>>>(x, partials, m, c, shift);
   bn_fwd_finalize_kernel<<<(c + 255) / 256, 256, 0, stream>>>(
       partials, nblk, c, m, eps, momentum, mean, invstd, running_mean,
-      running_var);
+      running_var, shift);
   int nblk2 = bn_nblocks(m * nq);
   size_t lds = (size_t)2 * c * sizeof(float);
   bn_fwd_norm_kernel<T><<<nblk2, kBnBlock, lds, stream>>>(
