@@ -402,7 +402,12 @@ static void bn_fwd_t(const T* x, T* y, long m, int c, float eps,
                      float* invstd, float* partials, hipStream_t stream) {
   const int nq = c / 4;
   const int nsub = kBnBlock / nq > 0 ? kBnBlock / nq : 1;
-  int nblk = bn_nblocks((m + nsub - 1) / nsub);
+  // One block per nsub-row stripe (capped): the parallel unit of the
+  // stats kernels is (row-stripe x quad), so the grid must scale with
+  // m/nsub, NOT m/nsub/256 (a /256 here left the chip <40% occupied).
+  long stripes = (m + nsub - 1) / nsub;
+  int nblk = stripes > kBnMaxBlocks ? kBnMaxBlocks
+                                    : (stripes < 1 ? 1 : (int)stripes);
   // The running mean doubles as the per-channel variance shift.
   const float* shift = running_mean;
   bn_fwd_partial_kernel<T>
@@ -423,7 +428,9 @@ static void bn_bwd_t(const T* dy, const T* x, T* dx, long m, int c,
                      float* partials, float* consts, hipStream_t stream) {
   const int nq = c / 4;
   const int nsub = kBnBlock / nq > 0 ? kBnBlock / nq : 1;
-  int nblk = bn_nblocks((m + nsub - 1) / nsub);
+  long stripes = (m + nsub - 1) / nsub;
+  int nblk = stripes > kBnMaxBlocks ? kBnMaxBlocks
+                                    : (stripes < 1 ? 1 : (int)stripes);
   bn_bwd_partial_kernel<T>
       <<<nblk, kBnBlock, 0, stream>>>(dy, x, partials, m, c, mean, invstd);
   float* ca = consts;
