@@ -157,27 +157,45 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
 }
 
 // Stage 2 (fwd): finalize mean/invstd, update running stats.
-__global__ void bn_fwd_finalize_kernel(
+// One BLOCK per channel: the nblk partial entries are reduced by a
+// 256-thread tree (a thread-per-channel serial loop over up to 1024
+// partials dominated the layer time).
+__global__ __launch_bounds__(kBnBlock) void bn_fwd_finalize_kernel(
     const float* __restrict__ partials, int nblk, int c, long m, float eps,
     float momentum, float* __restrict__ mean, float* __restrict__ invstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     const float* __restrict__ shift) {
-  int ch = blockIdx.x * blockDim.x + threadIdx.x;
-  if (ch >= c) return;
+  const int ch = blockIdx.x;
+  __shared__ float red[2][kBnBlock];
   float s = 0.f, q = 0.f;
-  for (int b = 0; b < nblk; ++b) {
+  for (int b = threadIdx.x; b < nblk; b += kBnBlock) {
     s += partials[((long)b * 2) * c + ch];
     q += partials[((long)b * 2 + 1) * c + ch];
   }
-  float d = s / (float)m;                     // mean of (x - shift)
-  float var = fmaxf(q / (float)m - d * d, 0.f);
-  float mu = d + (shift ? shift[ch] : 0.f);
-  mean[ch] = mu;
-  invstd[ch] = rsqrtf(var + eps);
-  if (running_mean != nullptr) {
-    running_mean[ch] = (1.f - momentum) * running_mean[ch] + momentum * mu;
-    float unbiased = (m > 1) ? var * (float)m / (float)(m - 1) : var;
-    running_var[ch] = (1.f - momentum) * running_var[ch] + momentum * unbiased;
+  red[0][threadIdx.x] = s;
+  red[1][threadIdx.x] = q;
+  __syncthreads();
+  for (int st = kBnBlock / 2; st > 0; st >>= 1) {
+    if ((int)threadIdx.x < st) {
+      red[0][threadIdx.x] += red[0][threadIdx.x + st];
+      red[1][threadIdx.x] += red[1][threadIdx.x + st];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    s = red[0][0];
+    q = red[1][0];
+    float d = s / (float)m;                   // mean of (x - shift)
+    float var = fmaxf(q / (float)m - d * d, 0.f);
+    float mu = d + (shift ? shift[ch] : 0.f);
+    mean[ch] = mu;
+    invstd[ch] = rsqrtf(var + eps);
+    if (running_mean != nullptr) {
+      running_mean[ch] = (1.f - momentum) * running_mean[ch] + momentum * mu;
+      float unbiased = (m > 1) ? var * (float)m / (float)(m - 1) : var;
+      running_var[ch] =
+          (1.f - momentum) * running_var[ch] + momentum * unbiased;
+    }
   }
 }
 
@@ -316,25 +334,39 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_partial_kernel(
 // Backward stage 2: finalize dbias/dscale -> param grads + dx constants.
 // dx = a*dy + b*xhat + cns  with a = w*invstd, b = -a*dscale/M,
 // cns = -a*dbias/M  (training-mode batch-norm gradient).
-__global__ void bn_bwd_finalize_kernel(
+__global__ __launch_bounds__(kBnBlock) void bn_bwd_finalize_kernel(
     const float* __restrict__ partials, int nblk, int c, long m,
     const float* __restrict__ weight, const float* __restrict__ invstd,
     float* __restrict__ dweight, float* __restrict__ dbias_out,
     float* __restrict__ ca, float* __restrict__ cb,
     float* __restrict__ cc) {
-  int ch = blockIdx.x * blockDim.x + threadIdx.x;
-  if (ch >= c) return;
+  const int ch = blockIdx.x;
+  __shared__ float red[2][kBnBlock];
   float db = 0.f, ds = 0.f;
-  for (int b = 0; b < nblk; ++b) {
+  for (int b = threadIdx.x; b < nblk; b += kBnBlock) {
     db += partials[((long)b * 2) * c + ch];
     ds += partials[((long)b * 2 + 1) * c + ch];
   }
-  dweight[ch] = ds;
-  dbias_out[ch] = db;
-  float a = weight[ch] * invstd[ch];
-  ca[ch] = a;
-  cb[ch] = -a * ds / (float)m;
-  cc[ch] = -a * db / (float)m;
+  red[0][threadIdx.x] = db;
+  red[1][threadIdx.x] = ds;
+  __syncthreads();
+  for (int st = kBnBlock / 2; st > 0; st >>= 1) {
+    if ((int)threadIdx.x < st) {
+      red[0][threadIdx.x] += red[0][threadIdx.x + st];
+      red[1][threadIdx.x] += red[1][threadIdx.x + st];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    db = red[0][0];
+    ds = red[1][0];
+    dweight[ch] = ds;
+    dbias_out[ch] = db;
+    float a = weight[ch] * invstd[ch];
+    ca[ch] = a;
+    cb[ch] = -a * ds / (float)m;
+    cc[ch] = -a * db / (float)m;
+  }
 }
 
 // Backward stage 3: dx elementwise.
@@ -412,7 +444,7 @@ static void bn_fwd_t(const T* x, T* y, long m, int c, float eps,
   const float* shift = running_mean;
   bn_fwd_partial_kernel<T>
       <<<nblk, kBnBlock, 0, stream>>>(x, partials, m, c, shift);
-  bn_fwd_finalize_kernel<<<(c + 255) / 256, 256, 0, stream>>>(
+  bn_fwd_finalize_kernel<<<c, kBnBlock, 0, stream>>>(
       partials, nblk, c, m, eps, momentum, mean, invstd, running_mean,
       running_var, shift);
   int nblk2 = bn_nblocks(m * nq);
@@ -436,7 +468,7 @@ static void bn_bwd_t(const T* dy, const T* x, T* dx, long m, int c,
   float* ca = consts;
   float* cb = consts + c;
   float* cc = consts + 2 * c;
-  bn_bwd_finalize_kernel<<<(c + 255) / 256, 256, 0, stream>>>(
+  bn_bwd_finalize_kernel<<<c, kBnBlock, 0, stream>>>(
       partials, nblk, c, m, weight, invstd, dweight, dbias, ca, cb, cc);
   int nblk2 = bn_nblocks(m * nq);
   size_t lds = (size_t)5 * c * sizeof(float);
