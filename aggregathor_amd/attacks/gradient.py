@@ -78,3 +78,56 @@ register("random", RandomAttack)
 register("zero", ZeroAttack)
 register("magnitude", MagnitudeAttack)
 register("nan", NaNAttack)
+
+
+class ALIEAttack(_Attack):
+    """'A Little Is Enough' (Baruch et al. 2019): submit mean + z * std of
+    the honest gradients, with z small enough to evade distance-based
+    selection yet consistently biased.
+
+    Omniscient variant: when the attacker can see other workers' gradients
+    (engine exposes the local honest rows via ``observe``), mean/std are
+    estimated from them; otherwise the worker's own gradient serves as the
+    mean estimate with zero std (degrading to a no-op).
+    """
+
+    graph_safe = True
+
+    def __init__(self, args):
+        self.args = tools.parse_keyval(args, defaults={"z": 1.0})
+        self._rows = None
+
+    def observe(self, honest_rows):
+        self._rows = honest_rows
+
+    def craft(self, honest, worker, step):
+        if self._rows is not None and self._rows.shape[0] >= 2:
+            mean = self._rows.mean(dim=0)
+            std = self._rows.std(dim=0, unbiased=False)
+            return mean + self.args["z"] * std
+        return honest
+
+
+class IPMAttack(_Attack):
+    """Inner-product manipulation (Xie et al. 2020): submit -eps * mean of
+    the honest gradients -- a negatively-aligned vector small enough to
+    survive distance filters. Omniscient like ALIE."""
+
+    graph_safe = True
+
+    def __init__(self, args):
+        self.args = tools.parse_keyval(args, defaults={"eps": 0.5})
+        self._rows = None
+
+    def observe(self, honest_rows):
+        self._rows = honest_rows
+
+    def craft(self, honest, worker, step):
+        base = (self._rows.mean(dim=0)
+                if self._rows is not None and self._rows.shape[0] >= 1
+                else honest)
+        return base * (-self.args["eps"])
+
+
+register("alie", ALIEAttack)
+register("ipm", IPMAttack)

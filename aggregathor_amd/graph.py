@@ -283,12 +283,31 @@ class Engine:
                 self._trace(f"worker {worker}: backward")
                 loss.backward()
                 losses.append(loss.detach())
-                # Byzantine replacement: the worker computed honestly, then lies.
-                if self.attack is not None and worker < self.nb_real_byz:
-                    self._trace(f"worker {worker}: byzantine craft")
-                    row.copy_(self.attack.craft(row.clone(), worker,
-                                                self.global_step))
+        self._apply_attack()
         return torch.stack(losses).mean()
+
+    def _apply_attack(self):
+        """Replace the local real-Byzantine workers' rows with crafted ones.
+
+        Runs AFTER all local gradients are computed so omniscient attacks
+        (ALIE/IPM) can observe this rank's honest rows. Byzantine worker ids
+        are the global prefix 0..nb_real_byz-1, so locally they are the
+        first rows and the honest remainder is a zero-copy slice.
+        """
+        if self.attack is None:
+            return
+        n_local_byz = sum(1 for w in self.group.worker_ids
+                          if w < self.nb_real_byz)
+        if n_local_byz == 0:
+            return
+        if hasattr(self.attack, "observe"):
+            self.attack.observe(self.local_rows[n_local_byz:])
+        for li in range(n_local_byz):
+            worker = self.group.worker_ids[li]
+            self._trace(f"worker {worker}: byzantine craft")
+            row = self.local_rows[li]
+            row.copy_(self.attack.craft(row.clone(), worker,
+                                        self.global_step))
 
     def aggregate(self, matrix=None):
         """Gather all rows (unless pre-gathered), verify integrity, inject

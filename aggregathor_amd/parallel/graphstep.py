@@ -133,9 +133,7 @@ class CapturedStep:
                         loss = loss.float() + eng._regularization()
                 loss.backward()
                 losses.append(loss.detach())
-                if eng.attack is not None and worker < eng.nb_real_byz:
-                    row.copy_(eng.attack.craft(row.clone(), worker,
-                                               eng.global_step))
+            eng._apply_attack()
         return torch.stack(losses).mean()
 
     def run(self):

@@ -187,3 +187,34 @@ def test_trace_phase_times(capsys):
     assert set(eng.phase_times) == {"local_gradients", "gather", "aggregate",
                                     "apply"}
     assert all(v > 0 for v in eng.phase_times.values())
+
+
+def test_alie_attack_krum_resists():
+    # ALIE with moderate z: krum f=2 should keep learning.
+    eng = _make_engine(aggregator="krum", n=8, f=2,
+                       nb_real_byz=2, attack="alie", attack_args=["z:1.5"])
+    for _ in range(120):
+        loss = eng.step()
+    assert math.isfinite(loss)
+    assert eng.evaluate()["top1-X-acc"] > 0.45
+
+
+def test_ipm_attack_median_resists():
+    eng = _make_engine(aggregator="averaged-median", n=8, f=2,
+                       nb_real_byz=2, attack="ipm", attack_args=["eps:0.5"])
+    for _ in range(120):
+        loss = eng.step()
+    assert math.isfinite(loss)
+    assert eng.evaluate()["top1-X-acc"] > 0.45
+
+
+def test_omniscient_attack_sees_honest_rows():
+    from aggregathor_amd import attacks as attacks_mod
+    eng = _make_engine(aggregator="average", n=4, f=1,
+                       nb_real_byz=1, attack="ipm", attack_args=["eps:1.0"])
+    eng.step()
+    # IPM row 0 must equal -eps * mean of rows 1..3 (the honest rows at the
+    # time of crafting).
+    got = eng.local_rows[0]
+    want = -1.0 * eng.local_rows[1:].mean(dim=0)
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-7)
