@@ -70,6 +70,12 @@ def main():
         n = 11 if gar == "bulyan" else 8
         configs.append(dict(gar=gar, n=n, f=2, attack="magnitude",
                             attack_args=["factor:1e6"]))
+    # Omniscient attacks: ALIE (mean + z*std) and IPM (-eps * mean).
+    for gar in ("average", "krum", "averaged-median"):
+        configs.append(dict(gar=gar, n=8, f=2, attack="alie",
+                            attack_args=["z:1.5"]))
+        configs.append(dict(gar=gar, n=8, f=2, attack="ipm",
+                            attack_args=["eps:1.0"]))
     # Data poisoning (mnistAttack severity 2, worker 0).
     for gar in ("average", "krum"):
         configs.append(dict(gar=gar, n=5, f=1, exp="mnistAttack",
