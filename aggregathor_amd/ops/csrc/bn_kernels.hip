@@ -49,10 +49,11 @@ __device__ __forceinline__ __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
   return __float2bfloat16(v);
 }
 
-// 4-channel vector of the data type (8 B for bf16, 16 B for fp32).
-template <typename T>
-struct alignas(4 * sizeof(T)) tvec4 {
-  T v[4];
+// W-channel vector of the data type (capped at 16 B alignment; a 32 B
+// fp32x8 struct is emitted as two dwordx4 accesses, which is fine).
+template <typename T, int W>
+struct alignas((W * sizeof(T)) < 16 ? (W * sizeof(T)) : 16) tvecw {
+  T v[W];
 };
 
 static inline int bn_nblocks(long total_quads) {
@@ -75,36 +76,36 @@ static inline int bn_nblocks(long total_quads) {
 // catastrophically once a channel's |mean| >> sigma; the running mean
 // tracks the batch mean after a few steps, so the shifted form stays
 // exact precisely when drift develops. shift == nullptr -> 0.
-template <typename T>
+template <typename T, int QW>
 __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
     const T* __restrict__ x, float* __restrict__ partials, long m, int c,
     const float* __restrict__ shift) {
-  const int nq = c / 4;
-  __shared__ float red[2][kBnBlock * 4];
+  const int nq = c / QW;
+  __shared__ float red[2][kBnBlock * 8];
 
   if (nq >= kBnBlock) {
     // Wide-channel regime (C >= 1024): one thread per quad, tiled over C.
     for (int q0 = 0; q0 < nq; q0 += kBnBlock) {
       int qt = q0 + (int)threadIdx.x;
       if (qt >= nq) break;
-      float sh[4];
+      float sh[QW];
 #pragma unroll
-      for (int k = 0; k < 4; ++k)
-        sh[k] = shift ? shift[qt * 4 + k] : 0.f;
-      float s[4] = {0.f, 0.f, 0.f, 0.f};
-      float q2[4] = {0.f, 0.f, 0.f, 0.f};
+      for (int k = 0; k < QW; ++k)
+        sh[k] = shift ? shift[qt * QW + k] : 0.f;
+      float s[QW] = {};
+      float q2[QW] = {};
       for (long row = blockIdx.x; row < m; row += gridDim.x) {
-        tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
+        tvecw<T, QW> v = reinterpret_cast<const tvecw<T, QW>*>(x + row * c)[qt];
 #pragma unroll
-        for (int k = 0; k < 4; ++k) {
+        for (int k = 0; k < QW; ++k) {
           float f = to_f32<T>(v.v[k]) - sh[k];
           s[k] += f;
           q2[k] = fmaf(f, f, q2[k]);
         }
       }
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        int ch = qt * 4 + k;
+      for (int k = 0; k < QW; ++k) {
+        int ch = qt * QW + k;
         partials[((long)blockIdx.x * 2) * c + ch] = s[k];
         partials[((long)blockIdx.x * 2 + 1) * c + ch] = q2[k];
       }
@@ -115,18 +116,18 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
   const int qt = (int)threadIdx.x % nq;
   const int sub = (int)threadIdx.x / nq;
   const int nsub = kBnBlock / nq;
-  float sh[4];
+  float sh[QW];
 #pragma unroll
-  for (int k = 0; k < 4; ++k) sh[k] = shift ? shift[qt * 4 + k] : 0.f;
-  float s[4] = {0.f, 0.f, 0.f, 0.f};
-  float q2[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int k = 0; k < QW; ++k) sh[k] = shift ? shift[qt * QW + k] : 0.f;
+  float s[QW] = {};
+  float q2[QW] = {};
   if (sub < nsub) {
     const long rows_per_grid = (long)gridDim.x * nsub;
     for (long row = (long)blockIdx.x * nsub + sub; row < m;
          row += rows_per_grid) {
-      tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
+      tvecw<T, QW> v = reinterpret_cast<const tvecw<T, QW>*>(x + row * c)[qt];
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
+      for (int k = 0; k < QW; ++k) {
         float f = to_f32<T>(v.v[k]) - sh[k];
         s[k] += f;
         q2[k] = fmaf(f, f, q2[k]);
@@ -134,22 +135,22 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
     }
   }
 #pragma unroll
-  for (int k = 0; k < 4; ++k) {
-    red[0][threadIdx.x * 4 + k] = s[k];
-    red[1][threadIdx.x * 4 + k] = q2[k];
+  for (int k = 0; k < QW; ++k) {
+    red[0][threadIdx.x * QW + k] = s[k];
+    red[1][threadIdx.x * QW + k] = q2[k];
   }
   __syncthreads();
   // Deterministic cross-sub combine: sub 0 of each quad sums in order.
   if (sub == 0) {
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
+    for (int k = 0; k < QW; ++k) {
       float ts = 0.f, tq = 0.f;
       for (int u = 0; u < nsub; ++u) {
         int t = u * nq + qt;
         ts += red[0][t * 4 + k];
         tq += red[1][t * 4 + k];
       }
-      int ch = qt * 4 + k;
+      int ch = qt * QW + k;
       partials[((long)blockIdx.x * 2) * c + ch] = ts;
       partials[((long)blockIdx.x * 2 + 1) * c + ch] = tq;
     }
@@ -200,7 +201,7 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_finalize_kernel(
 }
 
 // Stage 3 (fwd): y = (x - mean) * invstd * w + b, elementwise, vectorized.
-template <typename T>
+template <typename T, int QW>
 __global__ __launch_bounds__(kBnBlock) void bn_fwd_norm_kernel(
     const T* __restrict__ x, T* __restrict__ y, long m, int c,
     const float* __restrict__ mean, const float* __restrict__ invstd,
@@ -214,7 +215,7 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_norm_kernel(
     sb[ch] = bias[ch] - mean[ch] * a;
   }
   __syncthreads();
-  const int nq = c / 4;
+  const int nq = c / QW;
   const long stride = (long)gridDim.x * blockDim.x;
   // Incremental (row, quad) tracking: one 64-bit division per thread
   // instead of one per element (64-bit div is emulated and dominated the
@@ -225,14 +226,14 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_norm_kernel(
   const long dr = stride / nq;
   const int dq = (int)(stride - dr * nq);
   while (row < m) {
-    tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
-    tvec4<T> o;
+    tvecw<T, QW> v = reinterpret_cast<const tvecw<T, QW>*>(x + row * c)[qt];
+    tvecw<T, QW> o;
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
-      int ch = qt * 4 + k;
+    for (int k = 0; k < QW; ++k) {
+      int ch = qt * QW + k;
       o.v[k] = from_f32<T>(fmaf(to_f32<T>(v.v[k]), sa[ch], sb[ch]));
     }
-    reinterpret_cast<tvec4<T>*>(y + row * c)[qt] = o;
+    reinterpret_cast<tvecw<T, QW>*>(y + row * c)[qt] = o;
     qt += dq;
     row += dr;
     if (qt >= nq) {
@@ -245,29 +246,29 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_norm_kernel(
 // ---------------------------------------------------------------------------
 // Backward stage 1: per-block partial dbias = sum dy, dscale = sum dy*xhat.
 
-template <typename T>
+template <typename T, int QW>
 __global__ __launch_bounds__(kBnBlock) void bn_bwd_partial_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     float* __restrict__ partials, long m, int c,
     const float* __restrict__ mean, const float* __restrict__ invstd) {
-  const int nq = c / 4;
-  __shared__ float red[2][kBnBlock * 4];
+  const int nq = c / QW;
+  __shared__ float red[2][kBnBlock * 8];
 
   if (nq >= kBnBlock) {
     for (int q0 = 0; q0 < nq; q0 += kBnBlock) {
       int qt = q0 + (int)threadIdx.x;
       if (qt >= nq) break;
-      float mu[4], is[4], db[4] = {0, 0, 0, 0}, ds[4] = {0, 0, 0, 0};
+      float mu[QW], is[QW], db[QW] = {}, ds[QW] = {};
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        mu[k] = mean[qt * 4 + k];
-        is[k] = invstd[qt * 4 + k];
+      for (int k = 0; k < QW; ++k) {
+        mu[k] = mean[qt * QW + k];
+        is[k] = invstd[qt * QW + k];
       }
       for (long row = blockIdx.x; row < m; row += gridDim.x) {
-        tvec4<T> g = reinterpret_cast<const tvec4<T>*>(dy + row * c)[qt];
-        tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
+        tvecw<T, QW> g = reinterpret_cast<const tvecw<T, QW>*>(dy + row * c)[qt];
+        tvecw<T, QW> v = reinterpret_cast<const tvecw<T, QW>*>(x + row * c)[qt];
 #pragma unroll
-        for (int k = 0; k < 4; ++k) {
+        for (int k = 0; k < QW; ++k) {
           float gf = to_f32<T>(g.v[k]);
           float xh = (to_f32<T>(v.v[k]) - mu[k]) * is[k];
           db[k] += gf;
@@ -275,8 +276,8 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_partial_kernel(
         }
       }
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        int ch = qt * 4 + k;
+      for (int k = 0; k < QW; ++k) {
+        int ch = qt * QW + k;
         partials[((long)blockIdx.x * 2) * c + ch] = db[k];
         partials[((long)blockIdx.x * 2 + 1) * c + ch] = ds[k];
       }
@@ -286,22 +287,22 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_partial_kernel(
   const int qt = (int)threadIdx.x % nq;
   const int sub = (int)threadIdx.x / nq;
   const int nsub = kBnBlock / nq;
-  float mu[4], is[4];
+  float mu[QW], is[QW];
 #pragma unroll
-  for (int k = 0; k < 4; ++k) {
-    mu[k] = mean[qt * 4 + k];
-    is[k] = invstd[qt * 4 + k];
+  for (int k = 0; k < QW; ++k) {
+    mu[k] = mean[qt * QW + k];
+    is[k] = invstd[qt * QW + k];
   }
-  float db[4] = {0.f, 0.f, 0.f, 0.f};
-  float ds[4] = {0.f, 0.f, 0.f, 0.f};
+  float db[QW] = {};
+  float ds[QW] = {};
   if (sub < nsub) {
     const long rows_per_grid = (long)gridDim.x * nsub;
     for (long row = (long)blockIdx.x * nsub + sub; row < m;
          row += rows_per_grid) {
-      tvec4<T> g = reinterpret_cast<const tvec4<T>*>(dy + row * c)[qt];
-      tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
+      tvecw<T, QW> g = reinterpret_cast<const tvecw<T, QW>*>(dy + row * c)[qt];
+      tvecw<T, QW> v = reinterpret_cast<const tvecw<T, QW>*>(x + row * c)[qt];
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
+      for (int k = 0; k < QW; ++k) {
         float gf = to_f32<T>(g.v[k]);
         float xh = (to_f32<T>(v.v[k]) - mu[k]) * is[k];
         db[k] += gf;
@@ -310,21 +311,21 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_partial_kernel(
     }
   }
 #pragma unroll
-  for (int k = 0; k < 4; ++k) {
-    red[0][threadIdx.x * 4 + k] = db[k];
-    red[1][threadIdx.x * 4 + k] = ds[k];
+  for (int k = 0; k < QW; ++k) {
+    red[0][threadIdx.x * QW + k] = db[k];
+    red[1][threadIdx.x * QW + k] = ds[k];
   }
   __syncthreads();
   if (sub == 0) {
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
+    for (int k = 0; k < QW; ++k) {
       float tb = 0.f, tsc = 0.f;
       for (int u = 0; u < nsub; ++u) {
         int t = u * nq + qt;
         tb += red[0][t * 4 + k];
         tsc += red[1][t * 4 + k];
       }
-      int ch = qt * 4 + k;
+      int ch = qt * QW + k;
       partials[((long)blockIdx.x * 2) * c + ch] = tb;
       partials[((long)blockIdx.x * 2 + 1) * c + ch] = tsc;
     }
@@ -370,7 +371,7 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_finalize_kernel(
 }
 
 // Backward stage 3: dx elementwise.
-template <typename T>
+template <typename T, int QW>
 __global__ __launch_bounds__(kBnBlock) void bn_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
     long m, int c, const float* __restrict__ mean,
@@ -390,7 +391,7 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_dx_kernel(
     sc[ch] = cc[ch];
   }
   __syncthreads();
-  const int nq = c / 4;
+  const int nq = c / QW;
   const long stride = (long)gridDim.x * blockDim.x;
   long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long row = i0 / nq;
@@ -398,17 +399,17 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_dx_kernel(
   const long dr = stride / nq;
   const int dq = (int)(stride - dr * nq);
   while (row < m) {
-    tvec4<T> g = reinterpret_cast<const tvec4<T>*>(dy + row * c)[qt];
-    tvec4<T> v = reinterpret_cast<const tvec4<T>*>(x + row * c)[qt];
-    tvec4<T> o;
+    tvecw<T, QW> g = reinterpret_cast<const tvecw<T, QW>*>(dy + row * c)[qt];
+    tvecw<T, QW> v = reinterpret_cast<const tvecw<T, QW>*>(x + row * c)[qt];
+    tvecw<T, QW> o;
 #pragma unroll
-    for (int k = 0; k < 4; ++k) {
-      int ch = qt * 4 + k;
+    for (int k = 0; k < QW; ++k) {
+      int ch = qt * QW + k;
       float xh = (to_f32<T>(v.v[k]) - sm[ch]) * si[ch];
       float r = fmaf(to_f32<T>(g.v[k]), sa[ch], fmaf(xh, sb[ch], sc[ch]));
       o.v[k] = from_f32<T>(r);
     }
-    reinterpret_cast<tvec4<T>*>(dx + row * c)[qt] = o;
+    reinterpret_cast<tvecw<T, QW>*>(dx + row * c)[qt] = o;
     qt += dq;
     row += dr;
     if (qt >= nq) {
@@ -427,12 +428,12 @@ long bn_partials_elems(int c) {
   return (long)kBnMaxBlocks * 2 * c;
 }
 
-template <typename T>
+template <typename T, int QW>
 static void bn_fwd_t(const T* x, T* y, long m, int c, float eps,
                      float momentum, const float* weight, const float* bias,
                      float* running_mean, float* running_var, float* mean,
                      float* invstd, float* partials, hipStream_t stream) {
-  const int nq = c / 4;
+  const int nq = c / QW;
   const int nsub = kBnBlock / nq > 0 ? kBnBlock / nq : 1;
   // One block per nsub-row stripe (capped): the parallel unit of the
   // stats kernels is (row-stripe x quad), so the grid must scale with
@@ -442,28 +443,28 @@ static void bn_fwd_t(const T* x, T* y, long m, int c, float eps,
                                     : (stripes < 1 ? 1 : (int)stripes);
   // The running mean doubles as the per-channel variance shift.
   const float* shift = running_mean;
-  bn_fwd_partial_kernel<T>
+  bn_fwd_partial_kernel<T, QW>
       <<<nblk, kBnBlock, 0, stream>>>(x, partials, m, c, shift);
   bn_fwd_finalize_kernel<<<c, kBnBlock, 0, stream>>>(
       partials, nblk, c, m, eps, momentum, mean, invstd, running_mean,
       running_var, shift);
   int nblk2 = bn_nblocks(m * nq);
   size_t lds = (size_t)2 * c * sizeof(float);
-  bn_fwd_norm_kernel<T><<<nblk2, kBnBlock, lds, stream>>>(
+  bn_fwd_norm_kernel<T, QW><<<nblk2, kBnBlock, lds, stream>>>(
       x, y, m, c, mean, invstd, weight, bias);
 }
 
-template <typename T>
+template <typename T, int QW>
 static void bn_bwd_t(const T* dy, const T* x, T* dx, long m, int c,
                      const float* weight, const float* mean,
                      const float* invstd, float* dweight, float* dbias,
                      float* partials, float* consts, hipStream_t stream) {
-  const int nq = c / 4;
+  const int nq = c / QW;
   const int nsub = kBnBlock / nq > 0 ? kBnBlock / nq : 1;
   long stripes = (m + nsub - 1) / nsub;
   int nblk = stripes > kBnMaxBlocks ? kBnMaxBlocks
                                     : (stripes < 1 ? 1 : (int)stripes);
-  bn_bwd_partial_kernel<T>
+  bn_bwd_partial_kernel<T, QW>
       <<<nblk, kBnBlock, 0, stream>>>(dy, x, partials, m, c, mean, invstd);
   float* ca = consts;
   float* cb = consts + c;
@@ -472,7 +473,7 @@ static void bn_bwd_t(const T* dy, const T* x, T* dx, long m, int c,
       partials, nblk, c, m, weight, invstd, dweight, dbias, ca, cb, cc);
   int nblk2 = bn_nblocks(m * nq);
   size_t lds = (size_t)5 * c * sizeof(float);
-  bn_bwd_dx_kernel<T><<<nblk2, kBnBlock, lds, stream>>>(
+  bn_bwd_dx_kernel<T, QW><<<nblk2, kBnBlock, lds, stream>>>(
       dy, x, dx, m, c, mean, invstd, ca, cb, cc);
 }
 
@@ -480,29 +481,56 @@ void bn_fwd(const void* x, void* y, long m, int c, int dtype, float eps,
             float momentum, const float* weight, const float* bias,
             float* running_mean, float* running_var, float* mean,
             float* invstd, float* partials, hipStream_t stream) {
-  if (dtype == 1)
-    bn_fwd_t<__hip_bfloat16>((const __hip_bfloat16*)x, (__hip_bfloat16*)y, m,
-                             c, eps, momentum, weight, bias, running_mean,
-                             running_var, mean, invstd, partials, stream);
-  else
-    bn_fwd_t<float>((const float*)x, (float*)y, m, c, eps, momentum, weight,
-                    bias, running_mean, running_var, mean, invstd, partials,
-                    stream);
+  if (c % 8 == 0) {
+    if (dtype == 1)
+      bn_fwd_t<__hip_bfloat16, 8>((const __hip_bfloat16*)x,
+                                  (__hip_bfloat16*)y, m, c, eps, momentum,
+                                  weight, bias, running_mean, running_var,
+                                  mean, invstd, partials, stream);
+    else
+      bn_fwd_t<float, 8>((const float*)x, (float*)y, m, c, eps, momentum,
+                         weight, bias, running_mean, running_var, mean,
+                         invstd, partials, stream);
+  } else {
+    if (dtype == 1)
+      bn_fwd_t<__hip_bfloat16, 4>((const __hip_bfloat16*)x,
+                                  (__hip_bfloat16*)y, m, c, eps, momentum,
+                                  weight, bias, running_mean, running_var,
+                                  mean, invstd, partials, stream);
+    else
+      bn_fwd_t<float, 4>((const float*)x, (float*)y, m, c, eps, momentum,
+                         weight, bias, running_mean, running_var, mean,
+                         invstd, partials, stream);
+  }
 }
 
 void bn_bwd(const void* dy, const void* x, void* dx, long m, int c,
             int dtype, const float* weight, const float* mean,
             const float* invstd, float* dweight, float* dbias,
             float* partials, float* consts, hipStream_t stream) {
-  if (dtype == 1)
-    bn_bwd_t<__hip_bfloat16>((const __hip_bfloat16*)dy,
-                             (const __hip_bfloat16*)x, (__hip_bfloat16*)dx, m,
-                             c, weight, mean, invstd, dweight, dbias,
-                             partials, consts, stream);
-  else
-    bn_bwd_t<float>((const float*)dy, (const float*)x, (float*)dx, m, c,
-                    weight, mean, invstd, dweight, dbias, partials, consts,
-                    stream);
+  if (c % 8 == 0) {
+    if (dtype == 1)
+      bn_bwd_t<__hip_bfloat16, 8>((const __hip_bfloat16*)dy,
+                                  (const __hip_bfloat16*)x,
+                                  (__hip_bfloat16*)dx, m, c, weight, mean,
+                                  invstd, dweight, dbias, partials, consts,
+                                  stream);
+    else
+      bn_bwd_t<float, 8>((const float*)dy, (const float*)x, (float*)dx, m,
+                         c, weight, mean, invstd, dweight, dbias, partials,
+                         consts, stream);
+  } else {
+    if (dtype == 1)
+      bn_bwd_t<__hip_bfloat16, 4>((const __hip_bfloat16*)dy,
+                                  (const __hip_bfloat16*)x,
+                                  (__hip_bfloat16*)dx, m, c, weight, mean,
+                                  invstd, dweight, dbias, partials, consts,
+                                  stream);
+    else
+      bn_bwd_t<float, 4>((const float*)dy, (const float*)x, (float*)dx, m,
+                         c, weight, mean, invstd, dweight, dbias, partials,
+                         consts, stream);
+  }
 }
 
 }  // namespace gar
