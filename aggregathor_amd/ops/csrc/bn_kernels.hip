@@ -147,8 +147,8 @@ __global__ __launch_bounds__(kBnBlock) void bn_fwd_partial_kernel(
       float ts = 0.f, tq = 0.f;
       for (int u = 0; u < nsub; ++u) {
         int t = u * nq + qt;
-        ts += red[0][t * 4 + k];
-        tq += red[1][t * 4 + k];
+        ts += red[0][t * QW + k];
+        tq += red[1][t * QW + k];
       }
       int ch = qt * QW + k;
       partials[((long)blockIdx.x * 2) * c + ch] = ts;
@@ -322,8 +322,8 @@ __global__ __launch_bounds__(kBnBlock) void bn_bwd_partial_kernel(
       float tb = 0.f, tsc = 0.f;
       for (int u = 0; u < nsub; ++u) {
         int t = u * nq + qt;
-        tb += red[0][t * 4 + k];
-        tsc += red[1][t * 4 + k];
+        tb += red[0][t * QW + k];
+        tsc += red[1][t * QW + k];
       }
       int ch = qt * QW + k;
       partials[((long)blockIdx.x * 2) * c + ch] = tb;
