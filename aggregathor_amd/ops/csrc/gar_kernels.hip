@@ -59,11 +59,12 @@ static inline int nblocks_d(long d) {
 // dividing d (a float4 access on a row with d % 4 != 0 would be
 // misaligned). Launchers dispatch VW in {4, 2, 1}.
 template <int VW>
-struct alignas(VW * 4) fvec {
+struct alignas((VW * 4) < 16 ? (VW * 4) : 16) fvec {
   float v[VW];
 };
 
 static inline int vec_width(long d) {
+  if (d % 8 == 0) return 8;
   if (d % 4 == 0) return 4;
   if (d % 2 == 0) return 2;
   return 1;
@@ -593,7 +594,10 @@ void selection_average(const float* g, int n, long d, const int* sel, int m,
   (void)n;
   int nblk = nblocks_d(d);
   int vw = vec_width(d);
-  if (vw == 4)
+  if (vw == 8)
+    selection_average_kernel<8>
+        <<<nblk, kBlock, 0, stream>>>(g, out, d / 8, d, sel, m);
+  else if (vw == 4)
     selection_average_kernel<4>
         <<<nblk, kBlock, 0, stream>>>(g, out, d / 4, d, sel, m);
   else if (vw == 2)
@@ -832,6 +836,7 @@ static void launch_coordwise_n(const float* g, int n, long d, float* out,
                                hipStream_t stream, Args... args) {
   int nblk = nblocks_d(d);
   int vw = vec_width(d);
+  if (vw > 4) vw = 4;  // register balance: 8-wide doubles the column regs
   if (vw == 4)
     coordwise_kernel<NMAX, 4, OP<NMAX>>
         <<<nblk, kBlock, 0, stream>>>(g, out, d / 4, d, n, OP<NMAX>{args...});
@@ -950,6 +955,7 @@ void bulyan_final(const float* g, int n, long d, int f, int m,
   bulyan_inv_mk_kernel<<<1, 64, 0, stream>>>(inv_mk, t, m);
   int nblk = nblocks_d(d);
   int vw = vec_width(d);
+  if (vw > 4) vw = 4;  // register balance
   long dv = d / vw;
 #define GAR_BULYAN_LAUNCH(NMAX, VW)                                   \
   bulyan_final_kernel<NMAX, VW><<<nblk, kBlock, 0, stream>>>(         \
