@@ -71,7 +71,8 @@ class Trainer:
                  summary_period=config.default_summary_period,
                  evaluation_file=None, evaluation_delta=config.default_evaluation_delta,
                  evaluation_period=config.default_evaluation_period,
-                 rank0_only_services=True):
+                 rank0_only_services=True,
+                 profile_steps=0, profile_dir="profile_trace"):
         self.engine = engine
         self.max_step = max_step
         self.offstep = 0  # step at (re)start, for steps/s accounting
@@ -111,6 +112,12 @@ class Trainer:
                 self.summary_path = pathlib.Path(checkpoint_dir) / "summary.jsonl"
             if self.summary_path:
                 self.summary_path.parent.mkdir(parents=True, exist_ok=True)
+
+        # torch.profiler integration (per-phase profiler spans
+        # complement the rocprofv3 kernel evidence): profile `profile_steps`
+        # steps after a short warmup and export a chrome trace on rank 0.
+        self.profile_steps = profile_steps if self.is_rank0 else 0
+        self.profile_dir = profile_dir
 
         self._stop = threading.Event()
         self._threads = []
@@ -159,6 +166,16 @@ class Trainer:
     def train(self, progress_every=0):
         """Run to max_step; returns the perf report dict (runner.py:586-598)."""
         engine = self.engine
+        profiler = None
+        if self.profile_steps > 0:
+            import torch.profiler as tp
+            profiler = tp.profile(
+                activities=[tp.ProfilerActivity.CPU,
+                            tp.ProfilerActivity.CUDA],
+                schedule=tp.schedule(wait=1, warmup=2,
+                                     active=self.profile_steps, repeat=1),
+                on_trace_ready=tp.tensorboard_trace_handler(self.profile_dir))
+            profiler.__enter__()
         for t in self._threads:
             t.start()
         first_step_time = None
@@ -175,6 +192,8 @@ class Trainer:
                 if first_step_time is None:
                     first_step_time = dt
                 in_step += dt
+                if profiler is not None:
+                    profiler.step()
                 if progress_every and steps_done % progress_every == 0 and self.is_rank0:
                     tools.info(f"step {engine.global_step}  loss {loss:.5f}  "
                                f"({dt * 1e3:.1f} ms/step)")
@@ -185,6 +204,8 @@ class Trainer:
                                 f"{engine.global_step}; aborting")
                     break
         finally:
+            if profiler is not None:
+                profiler.__exit__(None, None, None)
             self._stop.set()
             for t in self._threads:
                 t.join(timeout=30.)

@@ -102,6 +102,10 @@ parser.add_argument("--trace", action="store_true", default=False,
 parser.add_argument("--amp", action="store_true", default=False,
                     help="bf16 autocast compute (fp32 gradients/aggregation)")
 parser.add_argument("--seed", type=int, default=1234)
+parser.add_argument("--profile-steps", type=int, default=0,
+                    help="Profile this many steps with torch.profiler and "
+                         "export a chrome trace (rank 0)")
+parser.add_argument("--profile-dir", type=str, default="profile_trace")
 parser.add_argument("--progress-every", type=int, default=100,
                     help="Print loss every K steps (0 = silent)")
 parser.add_argument("--stdout-to", type=str, default="-")
@@ -185,7 +189,8 @@ def main():
             summary_period=args.summary_period,
             evaluation_file=args.evaluation_file or None,
             evaluation_delta=args.evaluation_delta,
-            evaluation_period=args.evaluation_period)
+            evaluation_period=args.evaluation_period,
+            profile_steps=args.profile_steps, profile_dir=args.profile_dir)
         report = trainer.train(progress_every=args.progress_every)
         if report["diverged"]:
             sys.exit(1)

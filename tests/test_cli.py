@@ -102,3 +102,14 @@ def test_deploy_local_two_ranks(tmp_path):
     assert r.returncode == 0, (r.stdout.decode()[-1500:]
                                + r.stderr.decode()[-1500:])
     assert "all ranks completed" in r.stdout.decode()
+
+
+def test_runner_torch_profiler(tmp_path):
+    prof = tmp_path / "trace"
+    r = _run(["runner.py", "--experiment", "mnist", "--aggregator", "average",
+              "--nb-workers", "2", "--experiment-args", "batch-size:16",
+              "--max-step", "8", "--evaluation-delta", "-1",
+              "--evaluation-period", "-1", "--progress-every", "0",
+              "--profile-steps", "3", "--profile-dir", str(prof)])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    assert list(prof.glob("*.json*")), "no profiler trace exported"
