@@ -79,7 +79,6 @@ __host__ __device__ static inline int pair_index(int i, int j, int rows) {
 
 static inline int pair_layout_rows(int n) {
   if (n <= 8) return 8;
-  if (n <= 16) return 16;
   return n;  // tile kernel uses n-enumeration
 }
 
@@ -485,18 +484,16 @@ void sqdist(const float* g, int n, long d, float* partials, float* dist,
     else
       sqdist_small_kernel<8, 1>
           <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
-  } else if (n <= 16) {
-    if (vw >= 2)
-      sqdist_small_kernel<16, 2>
-          <<<nblk, kBlock, 0, stream>>>(g, partials, d / 2, d, n);
-    else
-      sqdist_small_kernel<16, 1>
-          <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
   } else {
+    // n > 8: 8x8 row tiles (64 accumulators; data re-read ~(tiles+1)/2
+    // times through L2), VW up to 4.
     int tiles_per_row = (n + 7) / 8;
     int ntiles = tiles_per_row * (tiles_per_row + 1) / 2;
     dim3 grid(nblk, ntiles);
-    if (vw >= 2)
+    if (vw >= 4)
+      sqdist_tile_kernel<4><<<grid, kBlock, 0, stream>>>(g, partials, d / 4,
+                                                         d, n, tiles_per_row);
+    else if (vw == 2)
       sqdist_tile_kernel<2><<<grid, kBlock, 0, stream>>>(g, partials, d / 2,
                                                          d, n, tiles_per_row);
     else
@@ -710,17 +707,17 @@ __device__ __forceinline__ float coord_median(const float (&vals)[NMAX],
   unsigned keys[NMAX];
 #pragma unroll
   for (int i = 0; i < NMAX; ++i) {
-    if (i >= n) break;
+    if (i >= n) continue;
     keys[i] = sort_key(vals[i]);
   }
   float out = 0.f;
 #pragma unroll
   for (int i = 0; i < NMAX; ++i) {
-    if (i >= n) break;
+    if (i >= n) continue;
     int rank = 0;
 #pragma unroll
     for (int j = 0; j < NMAX; ++j) {
-      if (j >= n) break;
+      if (j >= n) continue;
       if (j != i)
         rank += (keys[j] < keys[i]) | ((keys[j] == keys[i]) & (j < i));
     }
@@ -737,17 +734,17 @@ __device__ __forceinline__ float coord_averaged_median(
   unsigned dkeys[NMAX];
 #pragma unroll
   for (int i = 0; i < NMAX; ++i) {
-    if (i >= n) break;
+    if (i >= n) continue;
     dkeys[i] = sort_key(fabsf(vals[i] - zero));
   }
   float sum = 0.f;
 #pragma unroll
   for (int i = 0; i < NMAX; ++i) {
-    if (i >= n) break;
+    if (i >= n) continue;
     int rank = 0;
 #pragma unroll
     for (int j = 0; j < NMAX; ++j) {
-      if (j >= n) break;
+      if (j >= n) continue;
       if (j != i)
         rank += (dkeys[j] < dkeys[i]) | ((dkeys[j] == dkeys[i]) & (j < i));
     }
@@ -769,7 +766,7 @@ __global__ __launch_bounds__(kBlock) void coordwise_kernel(
     fvec<VW> colv[NMAX];
 #pragma unroll
     for (int i = 0; i < NMAX; ++i) {
-      if (i >= n) break;
+      if (i >= n) continue;
       colv[i] = reinterpret_cast<const fvec<VW>*>(g + (long)i * d)[x];
     }
     fvec<VW> res;
@@ -778,7 +775,7 @@ __global__ __launch_bounds__(kBlock) void coordwise_kernel(
       float vals[NMAX];
 #pragma unroll
       for (int i = 0; i < NMAX; ++i) {
-        if (i >= n) break;
+        if (i >= n) continue;
         vals[i] = colv[i].v[c];
       }
       res.v[c] = op(vals, n);
@@ -791,7 +788,7 @@ __global__ __launch_bounds__(kBlock) void coordwise_kernel(
     float vals[NMAX];
 #pragma unroll
     for (int i = 0; i < NMAX; ++i) {
-      if (i >= n) break;
+      if (i >= n) continue;
       vals[i] = g[(long)i * d + x];
     }
     out[x] = op(vals, n);
@@ -821,7 +818,7 @@ struct AverageNanOp {
     float count = 0.f;
 #pragma unroll
     for (int i = 0; i < NMAX; ++i) {
-      if (i >= n) break;
+      if (i >= n) continue;
       if (isfinite(vals[i])) {
         sum += vals[i];
         count += 1.f;
@@ -837,6 +834,7 @@ static void launch_coordwise_n(const float* g, int n, long d, float* out,
   int nblk = nblocks_d(d);
   int vw = vec_width(d);
   if (vw > 4) vw = 4;  // register balance: 8-wide doubles the column regs
+  if (NMAX >= 32 && vw > 2) vw = 2;  // NMAX=32 @ VW=4 spills to scratch
   if (vw == 4)
     coordwise_kernel<NMAX, 4, OP<NMAX>>
         <<<nblk, kBlock, 0, stream>>>(g, out, d / 4, d, n, OP<NMAX>{args...});
@@ -856,6 +854,8 @@ static void launch_coordwise(const float* g, int n, long d, float* out,
   else if (n <= 16)
     launch_coordwise_n<16, OP>(g, n, d, out, stream, args...);
   else
+    // NMAX=32 at VW=4 spills the 128-register column to scratch; VW=2
+    // keeps it resident (measured in kernel-resource-usage).
     launch_coordwise_n<kMaxNCoord, OP>(g, n, d, out, stream, args...);
 }
 
@@ -895,7 +895,7 @@ __global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
     fvec<VW> colv[NMAX];
 #pragma unroll
     for (int i = 0; i < NMAX; ++i) {
-      if (i >= n) break;
+      if (i >= n) continue;
       colv[i] = reinterpret_cast<const fvec<VW>*>(g + (long)i * d)[x];
     }
     fvec<VW> res;
@@ -904,11 +904,11 @@ __global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
       float inters[NMAX];
 #pragma unroll
       for (int k = 0; k < NMAX; ++k) {
-        if (k >= t) break;
+        if (k >= t) continue;
         float s = 0.f;
 #pragma unroll
         for (int i = 0; i < NMAX; ++i) {
-          if (i >= n) break;
+          if (i >= n) continue;
           if (fl[k * n + i]) s += colv[i].v[c];
         }
         inters[k] = s * inv[k];
@@ -923,11 +923,11 @@ __global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
     float inters[NMAX];
 #pragma unroll
     for (int k = 0; k < NMAX; ++k) {
-      if (k >= t) break;
+      if (k >= t) continue;
       float s = 0.f;
 #pragma unroll
       for (int i = 0; i < NMAX; ++i) {
-        if (i >= n) break;
+        if (i >= n) continue;
         if (fl[k * n + i]) s += g[(long)i * d + x];
       }
       inters[k] = s * inv[k];
@@ -956,6 +956,7 @@ void bulyan_final(const float* g, int n, long d, int f, int m,
   int nblk = nblocks_d(d);
   int vw = vec_width(d);
   if (vw > 4) vw = 4;  // register balance
+  if (n > 16 && vw > 2) vw = 2;  // NMAX=32 @ VW=4 spills to scratch
   long dv = d / vw;
 #define GAR_BULYAN_LAUNCH(NMAX, VW)                                   \
   bulyan_final_kernel<NMAX, VW><<<nblk, kBlock, 0, stream>>>(         \
