@@ -79,6 +79,7 @@ __host__ __device__ static inline int pair_index(int i, int j, int rows) {
 
 static inline int pair_layout_rows(int n) {
   if (n <= 8) return 8;
+  if (n <= 16) return 16;
   return n;  // tile kernel uses n-enumeration
 }
 
@@ -484,16 +485,19 @@ void sqdist(const float* g, int n, long d, float* partials, float* dist,
     else
       sqdist_small_kernel<8, 1>
           <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
+  } else if (n <= 16) {
+    if (vw >= 2)
+      sqdist_small_kernel<16, 2>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d / 2, d, n);
+    else
+      sqdist_small_kernel<16, 1>
+          <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
   } else {
-    // n > 8: 8x8 row tiles (64 accumulators; data re-read ~(tiles+1)/2
-    // times through L2), VW up to 4.
+    // measured: the 8x8 tile at VW=4 (170 VGPR + scratch) loses to VW=2.
     int tiles_per_row = (n + 7) / 8;
     int ntiles = tiles_per_row * (tiles_per_row + 1) / 2;
     dim3 grid(nblk, ntiles);
-    if (vw >= 4)
-      sqdist_tile_kernel<4><<<grid, kBlock, 0, stream>>>(g, partials, d / 4,
-                                                         d, n, tiles_per_row);
-    else if (vw == 2)
+    if (vw >= 2)
       sqdist_tile_kernel<2><<<grid, kBlock, 0, stream>>>(g, partials, d / 2,
                                                          d, n, tiles_per_row);
     else
