@@ -180,6 +180,16 @@ class Engine:
             self.model = self.model.to(memory_format=torch.channels_last)
         group.broadcast_model(self.model)
         self.params = [p for p in self.model.parameters() if p.requires_grad]
+        for p in self.params:
+            # Grad views cover [0, numel) through the param's strides; that
+            # requires a dense layout (contiguous or channels-last).
+            if not (p.is_contiguous() or
+                    (p.dim() == 4 and
+                     p.is_contiguous(memory_format=torch.channels_last))):
+                raise tools.UserException(
+                    f"parameter of shape {tuple(p.shape)} has a non-dense "
+                    "layout; the flattened-gradient views require dense "
+                    "parameters")
         self.d = flat_size(self.params)
 
         self.gar = aggregators_mod.instantiate(
