@@ -203,6 +203,14 @@ class Engine:
 
         self.phase_times = {}
         self._eval_model = None
+        # Serializes training steps against the service threads' device work
+        # (evaluate / state_dict): concurrent eval-model builds + MIOpen
+        # benchmark-find + forwards interleaving with hipGraph replays were
+        # observed to corrupt training (loss collapse ~20 steps after the
+        # first eval fire; one segfault). The reference evaluated on separate
+        # devices; on one GPU, serialization is the sound choice.
+        import threading
+        self.lock = threading.Lock()
         self.nb_real_byz = nb_real_byz
         self.attack = None
         if nb_real_byz > 0 and attack:
@@ -367,6 +375,10 @@ class Engine:
         synchronization -- benchmark hot loops use this; the NaN-divergence
         check then only sees the value when one is requested).
         """
+        with self.lock:
+            return self._step_locked(sync_loss)
+
+    def _step_locked(self, sync_loss):
         if self.use_graphs and self._graphstep is None \
                 and self.global_step >= self.graph_warmup:
             self._trace("engaging hipGraph step capture")
@@ -423,17 +435,23 @@ class Engine:
         BN running stats are the local rank's, exactly like the reference's
         eval replicas reading the PS variables concurrently with updates.
         """
-        if self._eval_model is None:
-            self._eval_model = self.experiment.model().to(self.device)
-        self._eval_model.load_state_dict(self.model.state_dict())
-        return self.experiment.accuracy(self._eval_model, self.device)
+        with self.lock:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            if self._eval_model is None:
+                self._eval_model = self.experiment.model().to(self.device)
+            self._eval_model.load_state_dict(self.model.state_dict())
+            return self.experiment.accuracy(self._eval_model, self.device)
 
     def state_dict(self):
-        return {
-            "step": self.global_step,
-            "model": self.model.state_dict(),
-            "optimizer": self.optimizer.state_dict(),
-        }
+        with self.lock:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            return {
+                "step": self.global_step,
+                "model": self.model.state_dict(),
+                "optimizer": self.optimizer.state_dict(),
+            }
 
     def load_state_dict(self, state):
         self.global_step = state["step"]
