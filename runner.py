@@ -193,7 +193,12 @@ def main():
             profile_steps=args.profile_steps, profile_dir=args.profile_dir)
         report = trainer.train(progress_every=args.progress_every)
         if report["diverged"]:
-            sys.exit(1)
+            # Skip the torch/HIP C++ teardown: after an aborted captured-graph
+            # session it intermittently calls std::terminate from a reaper
+            # thread (SIGABRT), clobbering the exit status the caller needs.
+            sys.stdout.flush()
+            sys.stderr.flush()
+            os._exit(1)
 
 
 if __name__ == "__main__":
