@@ -52,6 +52,9 @@ class CapturedStep:
 
     @staticmethod
     def supported(engine):
+        import os
+        if os.environ.get("AGGREGATHOR_NO_GRAPHS") == "1":
+            return False
         return (engine.device.type == "cuda"
                 and engine.lossy is None
                 and engine.integrity is None  # host-side MAC computation
