@@ -101,6 +101,11 @@ parser.add_argument("--trace", action="store_true", default=False,
                          "step execution")
 parser.add_argument("--amp", action="store_true", default=False,
                     help="bf16 autocast compute (fp32 gradients/aggregation)")
+parser.add_argument("--graphs", action="store_true", default=False,
+                    help="hipGraph step capture (opt-in here: long runs with "
+                         "the evaluation service showed divergence on some "
+                         "configs, under investigation -- see NOTES.md; "
+                         "bench.py uses graphs on its validated configs)")
 parser.add_argument("--seed", type=int, default=1234)
 parser.add_argument("--profile-steps", type=int, default=0,
                     help="Profile this many steps with torch.profiler and "
@@ -175,7 +180,8 @@ def main():
             l1_regularize=args.l1_regularize, l2_regularize=args.l2_regularize,
             nb_real_byz=args.nb_real_byz_workers, attack=args.attack,
             attack_args=args.attack_args or [], lossy=lossy, amp=args.amp,
-            trace=args.trace, seed=args.seed, integrity=integrity)
+            trace=args.trace, seed=args.seed, integrity=integrity,
+            use_graphs=("auto" if args.graphs else False))
         tools.info(f"model d = {engine.d} parameters, GAR = {args.aggregator}")
 
     with tools.Context("session", "info"):
