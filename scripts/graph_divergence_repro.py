@@ -69,8 +69,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--steps", type=int, default=400)
     ap.add_argument("--case", type=str, default="all",
-                    help="one of %s or 'all'" % ", ".join(CASES))
+                    help="one of %s, 'purity', or 'all'" % ", ".join(CASES))
     args = ap.parse_args()
+    if args.case == "purity":
+        replay_purity_check()
+        return
     names = list(CASES) if args.case == "all" else [args.case]
     results = {}
     for name in names:
@@ -82,3 +85,36 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def replay_purity_check(steps_before=5):
+    """Decisive test for the stateful-kernel hypothesis: freeze the inputs
+    and replay the captured local phase twice -- the produced gradient rows
+    must be BITWISE identical. A drift implicates a capture-unsafe kernel
+    (e.g. an MIOpen wrw solver accumulating into a lazily-zeroed workspace).
+    Candidate fix to A/B if drift is found: torch.backends.cudnn
+    .deterministic = True (excludes atomic-accumulation conv algorithms).
+    """
+    from aggregathor_amd import experiments
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    exp = experiments.instantiate(
+        "resnet50-cifar10", ["batch-size:32", "eval-examples:0"])
+    eng = Engine(exp, "average", WorkerGroup(8, device="cuda:0"),
+                 amp=True, use_graphs=True, graph_warmup=1,
+                 learning_rate_args=["initial-rate:0.0"])  # lr 0: params frozen
+    for _ in range(steps_before):
+        eng.step()
+    gs = eng._graphstep
+    assert gs is not None and gs.ready, "graphs did not engage"
+    # Stage once, replay the LOCAL graph twice on identical state.
+    gs._stage_batches()
+    gs.graph_local.replay()
+    torch.cuda.synchronize()
+    first = eng.local_rows.clone()
+    gs.graph_local.replay()
+    torch.cuda.synchronize()
+    drift = (eng.local_rows - first).abs().max().item()
+    same = torch.equal(eng.local_rows, first)
+    print(f"replay purity: bitwise_equal={same} max_drift={drift:.3e}")
+    return same
