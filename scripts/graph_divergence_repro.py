@@ -83,10 +83,6 @@ def main():
         print(f"{name:18s} {'diverged@%d' % bad if bad is not None else 'healthy'}")
 
 
-if __name__ == "__main__":
-    main()
-
-
 def replay_purity_check(steps_before=5):
     """Decisive test for the stateful-kernel hypothesis: freeze the inputs
     and replay the captured local phase twice -- the produced gradient rows
@@ -118,3 +114,7 @@ def replay_purity_check(steps_before=5):
     same = torch.equal(eng.local_rows, first)
     print(f"replay purity: bitwise_equal={same} max_drift={drift:.3e}")
     return same
+
+
+if __name__ == "__main__":
+    main()
