@@ -14,15 +14,17 @@ REQUIRED_KEYS = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
 
 
 def _clean_env():
-    return {k: v for k, v in os.environ.items()
-            if k not in ("RANK", "WORLD_SIZE", "LOCAL_RANK",
-                         "MASTER_ADDR", "MASTER_PORT")}
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE", "LOCAL_RANK",
+                        "MASTER_ADDR", "MASTER_PORT")}
+    env["OMP_NUM_THREADS"] = "4"  # avoid CPU thrash with multiple ranks
+    return env
 
 
 def test_bench_json_contract():
     r = subprocess.run(
-        [sys.executable, "bench.py", "--model", "resnet18", "--steps", "2",
-         "--warmup", "1", "--image-size", "32", "--batch-size", "2",
+        [sys.executable, "bench.py", "--model", "resnet20", "--dataset",
+         "cifar10", "--steps", "2", "--warmup", "1", "--batch-size", "2",
          "--device", "cpu"],
         capture_output=True, timeout=600, env=_clean_env(), cwd=str(REPO))
     assert r.returncode == 0, r.stderr.decode()[-2000:]
@@ -39,7 +41,7 @@ def test_bench_json_contract():
     cfg = d["config"]
     for k in ("model", "global_batch", "parallelism", "gar", "n_workers", "f"):
         assert k in cfg, k
-    assert cfg["model"] == "resnet18" and cfg["n_workers"] == 8 and cfg["f"] == 2
+    assert cfg["model"] == "resnet20" and cfg["n_workers"] == 8 and cfg["f"] == 2
     assert cfg["parallelism"] == "dp1"
 
 
@@ -47,9 +49,9 @@ def test_bench_under_torchrun_world2():
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29911", "bench.py", "--model", "resnet18",
-         "--gpus", "2", "--steps", "2", "--warmup", "1", "--image-size",
-         "32", "--batch-size", "2", "--device", "cpu"],
+         "--master-port", "29911", "bench.py", "--model", "resnet20",
+         "--dataset", "cifar10", "--gpus", "2", "--steps", "2", "--warmup",
+         "1", "--batch-size", "2", "--device", "cpu"],
         capture_output=True, timeout=900, env=_clean_env(), cwd=str(REPO))
     assert r.returncode == 0, r.stderr.decode()[-2000:]
     lines = [l for l in r.stdout.decode().splitlines() if l.startswith("{")]
