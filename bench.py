@@ -34,6 +34,10 @@ os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 import torch
 
 torch.backends.cudnn.benchmark = True
+if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
+    # Excludes atomic-accumulation conv algorithms (candidate fix for the
+    # hipGraph open issue, NOTES.md).
+    torch.backends.cudnn.deterministic = True
 
 
 def main():

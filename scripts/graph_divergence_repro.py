@@ -23,6 +23,10 @@ os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 import torch
 
 torch.backends.cudnn.benchmark = True
+if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
+    # Candidate fix for the capture-unsafe-solver hypothesis: deterministic
+    # mode excludes atomic-accumulation conv algorithms.
+    torch.backends.cudnn.deterministic = True
 
 
 def run_case(name, graphs, eval_every, l2, steps, seed=1234):
