@@ -147,10 +147,10 @@ def main():
         use_gpu = (args.use_gpu or args.reuse_gpu) and torch.cuda.is_available()
         if use_gpu:
             torch.backends.cudnn.benchmark = True
-if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
-    # Excludes atomic-accumulation conv algorithms (candidate fix for the
-    # hipGraph open issue, NOTES.md).
-    torch.backends.cudnn.deterministic = True
+            if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
+                # Excludes atomic-accumulation conv algorithms (candidate
+                # fix for the hipGraph open issue, NOTES.md).
+                torch.backends.cudnn.deterministic = True
         if (args.use_gpu or args.reuse_gpu) and not torch.cuda.is_available():
             tools.warning("--use-gpu requested but no GPU is available; "
                           "falling back to CPU")
