@@ -218,3 +218,14 @@ def test_omniscient_attack_sees_honest_rows():
     got = eng.local_rows[0]
     want = -1.0 * eng.local_rows[1:].mean(dim=0)
     torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-7)
+
+
+def test_cnnet_learns():
+    # CNNet needs a gentler lr than the MLP (dense-384 trunc-normal stack).
+    exp = experiments.instantiate("cnnet", ["batch-size:64"])
+    eng = Engine(exp, "median", WorkerGroup(4, device="cpu"), nbbyzwrks=1,
+                 learning_rate_args=["initial-rate:0.02"])
+    for _ in range(150):
+        loss = eng.step()
+    assert math.isfinite(loss)
+    assert eng.evaluate()["top1-X-acc"] > 0.3
