@@ -74,8 +74,11 @@ class Trainer:
                  rank0_only_services=True,
                  profile_steps=0, profile_dir="profile_trace"):
         self.engine = engine
+        # Reference semantics (runner.py --max-step help): the number of
+        # ADDITIONAL steps to perform -- on resume, training continues for
+        # max_step more steps past the restored global step.
         self.max_step = max_step
-        self.offstep = 0  # step at (re)start, for steps/s accounting
+        self.offstep = 0  # step at (re)start
         self.is_rank0 = engine.group.rank == 0
         services = self.is_rank0 or not rank0_only_services
 
@@ -183,8 +186,9 @@ class Trainer:
         steps_done = 0
         t_total0 = time.monotonic()
         diverged = False
+        target = self.offstep + self.max_step if self.max_step > 0 else -1
         try:
-            while self.max_step <= 0 or engine.global_step < self.max_step:
+            while target < 0 or engine.global_step < target:
                 t0 = time.monotonic()
                 loss = engine.step()
                 dt = time.monotonic() - t0

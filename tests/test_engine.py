@@ -127,14 +127,16 @@ def test_trainer_services(tmp_path):
     from aggregathor_amd import tools
     ckpt = tools.Checkpoints(tmp_path / "ckpt")
     assert ckpt.can_restore()
-    # Resume continues the global step.
+    # Resume continues the global step; max_step counts ADDITIONAL steps
+    # (reference runner.py --max-step semantics).
     eng2 = _make_engine()
-    trainer2 = Trainer(eng2, max_step=30, checkpoint_dir=str(tmp_path / "ckpt"),
+    trainer2 = Trainer(eng2, max_step=5, checkpoint_dir=str(tmp_path / "ckpt"),
                        checkpoint_delta=-1, checkpoint_period=-1,
                        evaluation_delta=-1, evaluation_period=-1)
-    assert eng2.global_step > 0  # restored
+    restored_at = eng2.global_step
+    assert restored_at > 0  # restored
     report2 = trainer2.train()
-    assert eng2.global_step == 30
+    assert eng2.global_step == restored_at + 5
 
 
 def test_nan_divergence_abort():
