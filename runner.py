@@ -74,7 +74,9 @@ parser.add_argument("--integrity-key", type=str, default="",
                     help="Enable per-step gradient integrity MACs with this "
                          "shared secret (the reference's message-signing "
                          "equivalent; corrupted rows are NaN-filled)")
-parser.add_argument("--max-step", type=int, default=config.default_max_step)
+parser.add_argument("--max-step", type=int, default=config.default_max_step,
+                    help="Number of additional steps to perform before "
+                         "stopping the training, non-positive for no limit")
 parser.add_argument("--checkpoint-dir", type=str, default="")
 parser.add_argument("--checkpoint-delta", type=int,
                     default=config.default_checkpoint_delta)
