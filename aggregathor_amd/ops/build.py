@@ -1,5 +1,10 @@
 """In-tree build of the ``_gar_hip`` torch extension for gfx950.
 
+(The reference's auto-compiling native-op infrastructure equivalent,
+/root/reference/native/__init__.py:208-304: mtime-incremental compilation of
+the native op library at import/build time; here a single hipcc invocation
+produces one in-tree .so.)
+
 Direct hipcc invocation (no hipify, no CUDA compatibility machinery): the
 kernels are native HIP/CDNA4 and the glue uses the c10::hip API directly.
 The resulting ``_gar_hip.so`` is written next to this file so it ships with

@@ -8,7 +8,8 @@
 //
 // - The [n, d] gradient matrix (n <= 64 workers, d up to ~100M fp32) lives in
 //   HBM3E; every kernel is memory-bound, so the design minimizes passes over
-//   the matrix and vectorizes all global accesses as float4 (16 B/lane).
+//   the matrix and vectorizes global accesses (up to fvec<8> = 32 B/row per
+//   thread, width-dispatched on the row alignment that d admits).
 // - Pairwise distances: one pass over the matrix; each block accumulates ALL
 //   pair partial sums for its d-chunk in registers (each loaded element is
 //   reused n-1 times), then a deterministic fixed-shape tree reduction
@@ -57,7 +58,8 @@ static inline int nblocks_d(long d) {
 // Aligned fp32 vector: rows of the [n, d] matrix start at arbitrary
 // d-multiples, so the usable vector width is the largest power of two
 // dividing d (a float4 access on a row with d % 4 != 0 would be
-// misaligned). Launchers dispatch VW in {4, 2, 1}.
+// misaligned). Launchers dispatch VW in {8, 4, 2, 1} per kernel (wider
+// isn't always better: some NMAX variants clamp VW for register balance).
 template <int VW>
 struct alignas((VW * 4) < 16 ? (VW * 4) : 16) fvec {
   float v[VW];
