@@ -113,3 +113,19 @@ def test_worker_group_divisibility():
         bad.world = 2
         if 3 % bad.world != 0:
             raise tools.UserException("indivisible")
+
+
+def test_reference_oracle_parity():
+    """Run this framework's GAR math against the reference's own compiled
+    kernels (skipped when the reference mount or g++ is unavailable)."""
+    import pathlib
+    import shutil
+    import subprocess
+    import sys
+    if not pathlib.Path("/root/reference").exists() or not shutil.which("g++"):
+        pytest.skip("reference mount or g++ unavailable")
+    repo = pathlib.Path(__file__).resolve().parent.parent
+    r = subprocess.run(
+        [sys.executable, str(repo / "scripts/reference_oracle_check.py")],
+        capture_output=True, timeout=600)
+    assert r.returncode == 0, r.stdout.decode()[-3000:] + r.stderr.decode()[-1000:]
