@@ -113,3 +113,23 @@ def test_runner_torch_profiler(tmp_path):
               "--profile-steps", "3", "--profile-dir", str(prof)])
     assert r.returncode == 0, r.stderr.decode()[-2000:]
     assert list(prof.glob("*.json*")), "no profiler trace exported"
+
+
+def test_reference_readme_loopback_invocation(tmp_path):
+    """The reference README's documented local-deployment command line
+    (reference README.md:146) runs verbatim (with max-step shortened):
+    every flag is accepted with the same spelling and semantics."""
+    r = _run(["runner.py",
+              "--server", '{"local": ["127.0.0.1:7000"]}',
+              "--ps-job-name", "local", "--wk-job-name", "local",
+              "--ev-job-name", "local",
+              "--experiment", "mnist",
+              "--learning-rate-args", "initial-rate:0.05",
+              "--aggregator", "average", "--nb-workers", "4",
+              "--reuse-gpu", "--max-step", "10",
+              "--evaluation-period", "-1", "--checkpoint-period", "-1",
+              "--summary-period", "-1", "--evaluation-delta", "100",
+              "--checkpoint-delta", "-1", "--summary-delta", "-1",
+              "--no-wait", "--progress-every", "0"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    assert "Steps/s" in r.stdout.decode()
