@@ -6,6 +6,7 @@
 set -u
 
 RESULTS="${1:-results}"
+MAX_STEP="${MAX_STEP:-2000}"   # override for quick smoke runs
 mkdir -p "$RESULTS"
 
 PIDS=()
@@ -40,27 +41,27 @@ fi
 # tractable default; override by editing.)
 run mnist-average-n2 \
   --experiment mnist --experiment-args batch-size:50 \
-  --aggregator average --nb-workers 2 --max-step 2000 $GPU_FLAGS
+  --aggregator average --nb-workers 2 --max-step $MAX_STEP $GPU_FLAGS
 
 # Robust GARs under data poisoning (mnistAttack, worker 0 poisoned).
 run mnistAttack-krum-n5-f1 \
   --experiment mnistAttack --experiment-args batch-size:50 malformed-severity:2 \
-  --aggregator krum --nb-workers 5 --nb-decl-byz-workers 1 --max-step 2000 $GPU_FLAGS
+  --aggregator krum --nb-workers 5 --nb-decl-byz-workers 1 --max-step $MAX_STEP $GPU_FLAGS
 
 run mnistAttack-average-n5 \
   --experiment mnistAttack --experiment-args batch-size:50 malformed-severity:2 \
-  --aggregator average --nb-workers 5 --max-step 2000 $GPU_FLAGS
+  --aggregator average --nb-workers 5 --max-step $MAX_STEP $GPU_FLAGS
 
 # Gradient-reversal attack vs Multi-Krum (BASELINE.json config shape).
 run mnist-krum-reversal-n8-f2 \
   --experiment mnist --experiment-args batch-size:50 \
   --aggregator krum --nb-workers 8 --nb-decl-byz-workers 2 \
-  --nb-real-byz-workers 2 --attack reversal --max-step 2000 $GPU_FLAGS
+  --nb-real-byz-workers 2 --attack reversal --max-step $MAX_STEP $GPU_FLAGS
 
 # Lossy-transport (UDP-semantics) + NaN-tolerant GAR.
 run mnist-avgnan-lossy-n4 \
   --experiment mnist --experiment-args batch-size:50 \
   --aggregator average-nan --nb-workers 4 \
-  --lossy drop-rate:0.05 workers:0 --max-step 2000 $GPU_FLAGS
+  --lossy drop-rate:0.05 workers:0 --max-step $MAX_STEP $GPU_FLAGS
 
 echo "[experiments] all done; results in $RESULTS/"

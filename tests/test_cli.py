@@ -133,3 +133,32 @@ def test_reference_readme_loopback_invocation(tmp_path):
               "--no-wait", "--progress-every", "0"])
     assert r.returncode == 0, r.stderr.decode()[-2000:]
     assert "Steps/s" in r.stdout.decode()
+
+
+def test_experiments_sh_smoke(tmp_path):
+    """The experiment batch driver runs end to end (shortened steps)."""
+    env = _clean = {k: v for k, v in os.environ.items()
+                    if k not in ("RANK", "WORLD_SIZE", "LOCAL_RANK")}
+    env["MAX_STEP"] = "3"
+    r = subprocess.run(["bash", "experiments.sh", str(tmp_path / "out")],
+                       capture_output=True, timeout=600, env=env,
+                       cwd=str(REPO))
+    assert r.returncode == 0, r.stderr.decode()[-1500:]
+    out = r.stdout.decode()
+    assert "all done" in out
+    # Every run produced its redirected stdout log and checkpoint dir.
+    logs = list((tmp_path / "out").glob("*.stdout"))
+    assert len(logs) == 5, [p.name for p in (tmp_path / "out").iterdir()]
+
+
+def test_runner_trace_phases(tmp_path):
+    r = _run(["runner.py", "--experiment", "mnist", "--aggregator", "median",
+              "--nb-workers", "3", "--experiment-args", "batch-size:16",
+              "--max-step", "3", "--evaluation-delta", "-1",
+              "--evaluation-period", "-1", "--progress-every", "0",
+              "--trace"])
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    out = r.stdout.decode()
+    for marker in ("forward", "backward", "gather", "aggregate", "apply",
+                   "Phase local_gradients"):
+        assert marker in out, marker
