@@ -64,3 +64,16 @@ def test_experiment_train_batch_device():
     m = exp.model()
     loss = exp.loss(m, (x, y))
     assert loss.dim() == 0
+
+
+@pytest.mark.parametrize("exp_name", ["vgg11-imagenet", "mobilenet_v2-imagenet"])
+def test_extra_families_through_engine(exp_name):
+    """VGG/MobileNet families run through the full engine pipeline."""
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    exp = experiments.instantiate(
+        exp_name, ["batch-size:2", "image-size:32", "eval-examples:0"])
+    eng = Engine(exp, "krum", WorkerGroup(4, device="cpu"), nbbyzwrks=0)
+    for _ in range(2):
+        loss = eng.step()
+    assert loss == loss  # finite
