@@ -19,7 +19,7 @@ class SyntheticClassification:
     """Deterministic synthetic (inputs, labels) source."""
 
     def __init__(self, shape, classes, seed=1234, teacher=None,
-                 eval_examples=1024, scale=1.0, signal=0.5):
+                 eval_examples=1024, scale=1.0, signal=0.5, pool_size=8):
         """
         Args:
           shape:   per-example input shape, e.g. (784,) or (3, 32, 32)
@@ -32,6 +32,12 @@ class SyntheticClassification:
                    False = uniform random labels (throughput benchmarking).
           signal:  class-pattern amplitude (task difficulty knob)
           eval_examples: size of the held-out deterministic eval set
+          pool_size: GPU batch-pool size (see below); 0 disables pooling --
+                   every (worker, step) batch is freshly generated and
+                   identical to the CPU stream. Use 0 (or a large pool) for
+                   accuracy / attack-convergence evaluations; the small
+                   default pool is for throughput benchmarking, where
+                   per-step host RNG + H2D would bottleneck the MI355X step.
         """
         self.shape = tuple(shape)
         self.classes = classes
@@ -54,7 +60,8 @@ class SyntheticClassification:
         # (batch(worker, step) == cpu batch(worker, step % pool_size)) --
         # generating 20 MB of fresh host randoms per micro-batch would
         # bottleneck an MI355X training step on the host RNG + H2D copy.
-        self.pool_size = 8
+        # pool_size=0 disables pooling (GPU stream == CPU stream).
+        self.pool_size = pool_size
         self._pools = {}
 
     def _make(self, batch_size, gen):
@@ -75,7 +82,7 @@ class SyntheticClassification:
     def batch(self, batch_size, worker, step, device="cpu"):
         """Training batch for (worker, step): pure function of the seed."""
         device = torch.device(device)
-        if device.type == "cuda":
+        if device.type == "cuda" and self.pool_size > 0:
             key = (batch_size, worker, str(device))
             pool = self._pools.get(key)
             if pool is None:

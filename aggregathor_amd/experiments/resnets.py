@@ -11,6 +11,7 @@ configs are synthetic/random-init by definition).
 
 from . import _Experiment, register
 from .data import SyntheticClassification
+from .datasets import RealDataset
 from .. import tools
 from ..models import NETWORKS
 
@@ -24,7 +25,8 @@ class ResNetExperiment(_Experiment):
     def __init__(self, args, arch="resnet50", dataset="imagenet"):
         args = tools.parse_keyval(args, defaults={
             "batch-size": 32, "eval-batch-size": 256, "seed": 1234,
-            "eval-examples": 512, "image-size": 0})
+            "eval-examples": 512, "image-size": 0, "data-dir": "",
+            "data-pool": 8})
         if args["batch-size"] <= 0:
             raise tools.UserException("Cannot make batches of non-positive size")
         self.args = args
@@ -34,18 +36,31 @@ class ResNetExperiment(_Experiment):
         if args["image-size"] > 0:
             shape = (shape[0], args["image-size"], args["image-size"])
         self.classes = spec["classes"]
+        self._real = None
+        if args["data-dir"]:
+            # cifar10: the binary distribution; imagenet: npz tensor shards
+            # (no JPEG decoder in this image -- datasets.py docstring).
+            if dataset == "cifar10":
+                self._real = RealDataset.cifar10(args["data-dir"],
+                                                 seed=args["seed"])
+            else:
+                self._real = RealDataset.tensor_folder(args["data-dir"],
+                                                       seed=args["seed"])
         self._synth = SyntheticClassification(
             shape, self.classes, seed=args["seed"],
-            eval_examples=args["eval-examples"])
+            eval_examples=args["eval-examples"],
+            pool_size=args["data-pool"])
+
+    def train_batch(self, worker, step, device):
+        src = self._real if self._real is not None else self._synth
+        return src.batch(self.args["batch-size"], worker, step, device)
 
     def model(self):
         return NETWORKS[self.arch](num_classes=self.classes)
 
-    def train_batch(self, worker, step, device):
-        return self._synth.batch(self.args["batch-size"], worker, step, device)
-
     def eval_batches(self, device):
-        yield from self._synth.eval_batches(self.args["eval-batch-size"], device)
+        src = self._real if self._real is not None else self._synth
+        yield from src.eval_batches(self.args["eval-batch-size"], device)
 
 
 def _make(arch, dataset):

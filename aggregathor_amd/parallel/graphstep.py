@@ -136,7 +136,10 @@ class CapturedStep:
                         loss = loss.float() + eng._regularization()
                 loss.backward()
                 losses.append(loss.detach())
-            eng._apply_attack()
+        # Outside the autocast region, exactly like the eager path
+        # (graph.py compute_local_gradients): an attack crafting with
+        # autocast-eligible ops must see identical numerics either way.
+        eng._apply_attack()
         return torch.stack(losses).mean()
 
     def run(self):

@@ -104,3 +104,19 @@ class WorkerGroup:
         t = torch.tensor([value], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         return t.item()
+
+    def any_rank(self, flag):
+        """True iff `flag` is True on ANY rank (MAX all-reduce of a bit).
+
+        Used for the collective divergence abort: a rank whose local loss
+        goes non-finite must not exit alone -- the others would block in the
+        next ``gather`` until the process-group timeout. Every rank folds
+        its local flag in every step so all ranks agree on the abort step
+        (local losses differ per rank, so local detection steps would too).
+        """
+        if self.world == 1:
+            return bool(flag)
+        dev = self.device if self.backend == "nccl" else "cpu"
+        t = torch.tensor([1.0 if flag else 0.0], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return bool(t.item())

@@ -201,11 +201,19 @@ class Trainer:
                 if progress_every and steps_done % progress_every == 0 and self.is_rank0:
                     tools.info(f"step {engine.global_step}  loss {loss:.5f}  "
                                f"({dt * 1e3:.1f} ms/step)")
-                # NaN-divergence abort (runner.py:570-574).
-                if not math.isfinite(loss):
+                # NaN-divergence abort (runner.py:570-574). In distributed
+                # mode the abort is COLLECTIVE: local losses differ per
+                # rank, so a rank-local break would leave the other ranks
+                # blocked in the next all-gather until the process-group
+                # timeout -- every rank folds its flag in every step and all
+                # abort together at the same step.
+                local_bad = not math.isfinite(loss)
+                if engine.group.any_rank(local_bad):
                     diverged = True
-                    tools.error(f"Training diverged (loss = {loss}) at step "
-                                f"{engine.global_step}; aborting")
+                    detail = f"loss = {loss}" if local_bad else \
+                        "a peer rank's loss went non-finite"
+                    tools.error(f"Training diverged ({detail}) at step "
+                                f"{engine.global_step}; aborting on all ranks")
                     break
         finally:
             if profiler is not None:

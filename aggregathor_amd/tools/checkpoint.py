@@ -56,14 +56,20 @@ class Checkpoints:
         return int(stem[stem.rindex("-") + 1:])
 
     def restore(self, path=None, map_location="cpu"):
-        """Load a checkpoint payload (latest if ``path`` is None)."""
+        """Load a checkpoint payload (latest if ``path`` is None).
+
+        ``weights_only=True``: Trainer auto-restores the latest file at
+        startup, so an unpickle here would turn a tampered checkpoint dir
+        into code execution. Payloads are plain dicts of tensors/ints
+        (Engine.state_dict), which the weights-only unpickler accepts.
+        """
         from . import UserException
         self._update()
         if path is None:
             if not self.__available:
                 raise UserException("No storage file to restore")
             path = self.__available[-1]
-        return torch.load(str(path), map_location=map_location, weights_only=False)
+        return torch.load(str(path), map_location=map_location, weights_only=True)
 
     def save(self, payload, step):
         """Atomically save a checkpoint payload at the given step."""

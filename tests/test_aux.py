@@ -129,3 +129,19 @@ def test_reference_oracle_parity():
         [sys.executable, str(repo / "scripts/reference_oracle_check.py")],
         capture_output=True, timeout=600)
     assert r.returncode == 0, r.stdout.decode()[-3000:] + r.stderr.decode()[-1000:]
+
+
+def test_integrity_detects_single_element_sparse_corruption():
+    # The sampled MAC alone would catch a 1-element flip with probability
+    # ~sample/d; the full-row bit-checksum folded into the MAC catches it
+    # ALWAYS (int64 sum of float32 bit patterns changes for any single flip).
+    gi = GradientIntegrity("secret", nbworkers=2, sample=64)  # tiny sample
+    d = 100000
+    rows = torch.randn(2, d)
+    macs = gi.sign_rows(rows, [0, 1], step=11)
+    for trial in range(5):  # several positions, incl. ones never sampled
+        matrix = rows.clone()
+        pos = (trial * 31337 + 17) % d
+        matrix[1, pos] += 1e-6  # smallest interesting corruption
+        failed = gi.verify_matrix(matrix, macs, step=11)
+        assert failed == [1], f"missed corruption at {pos}"

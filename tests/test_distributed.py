@@ -82,3 +82,39 @@ def test_world4_bulyan(tmp_path):
     base = outs[0]["flat"]
     for o in outs[1:]:
         assert torch.equal(base, o["flat"])
+
+
+NAN_WORKER = REPO / "tests" / "dist_nan_worker.py"
+
+
+def test_world2_collective_nan_abort(tmp_path):
+    # Rank 1's loss goes non-finite at step 3: BOTH ranks must exit cleanly
+    # at the same step (collective abort), not block in the next all-gather
+    # until the 300 s process-group timeout.
+    import json
+    import time
+    world = 2
+    procs = []
+    t0 = time.monotonic()
+    for rank in range(world):
+        env = dict(os.environ)
+        env.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29685",
+            "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME", "lo"),
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, str(NAN_WORKER), "1", "3"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    metas = []
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=120)
+        assert p.returncode == 0, f"worker failed:\n{stderr.decode()[-2000:]}"
+        metas.append(json.loads(stdout.decode().strip().splitlines()[-1]))
+    wall = time.monotonic() - t0
+    assert all(m["diverged"] for m in metas)
+    # Same abort step on every rank (steps 0..2 healthy, abort during step 3).
+    assert metas[0]["steps"] == metas[1]["steps"] == 4
+    # Clean exit well under the process-group timeout.
+    assert wall < 60, f"abort took {wall:.1f}s -- ranks likely hung in gather"
