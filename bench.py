@@ -10,14 +10,23 @@ single process when N=1). The total worker count stays n=8 at every N (the
 GAR's n/f are properties of the training run, not the GPU count), so per-GPU
 work shrinks as N grows: STRONG scaling, fixed global batch.
 
-Synthetic ImageNet-shaped data (3x224x224, 1000 classes), random-init
-weights, bf16 autocast compute with fp32 gradients/aggregation/optimizer.
+BASELINE.json's metric is "Krum & Bulyan vs average": `--gar all` runs the
+three GARs back-to-back in one process and prints one contract JSON line per
+GAR (krum f=2, bulyan f=1 -- Bulyan requires n >= 4f+3, so f=2 would need
+n >= 11, indivisible by the GPU counts; keeping n=8 makes the three numbers
+directly comparable -- and average f=0). The default (no --gar flag) stays
+the single headline line: krum, n=8, f=2.
 
-Rank 0 prints exactly one JSON line with the whole-job steps/sec (max-over-
-ranks timing, barrier+synchronize bracketed).
+Synthetic ImageNet-shaped data (3x224x224, 1000 classes), random-init
+weights, bf16 autocast compute with fp32 gradients/aggregation/optimizer
+(`--no-amp` for fp32 end-to-end).
+
+Rank 0 prints exactly one JSON line per benchmarked GAR with the whole-job
+steps/sec (max-over-ranks timing, barrier+synchronize bracketed).
 """
 
 import argparse
+import gc
 import json
 import os
 import sys
@@ -40,52 +49,12 @@ if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
     torch.backends.cudnn.deterministic = True
 
 
-def main():
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--model", type=str, default="resnet50")
-    ap.add_argument("--dataset", type=str, default="imagenet",
-                    help="dataset shape: imagenet | cifar10")
-    ap.add_argument("--workers", type=int, default=8, help="total GAR workers n")
-    ap.add_argument("--f", type=int, default=2, help="declared Byzantine workers")
-    ap.add_argument("--gar", type=str, default="krum",
-                    help="aggregation rule (krum | bulyan | average | ...)")
-    ap.add_argument("--batch-size", type=int, default=32, help="per-worker batch")
-    ap.add_argument("--image-size", type=int, default=224)
-    ap.add_argument("--attack", type=str, default="",
-                    help="optional attack (e.g. reversal) mounted by f workers")
-    ap.add_argument("--lossy", nargs="*", default=None,
-                    help="UDP-style lossy-channel injection key:value args")
-    ap.add_argument("--device", type=str, default="",
-                    help="override device (debug; cpu allowed)")
-    ap.add_argument("--no-amp", action="store_true",
-                    help="disable bf16 autocast (fp32 compute)")
-    args = ap.parse_args()
-
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    rank = int(os.environ.get("RANK", "0"))
-    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    n_gpus = max(args.gpus, world)
-
-    if args.device:
-        device = args.device
-    elif torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        device = f"cuda:{local_rank}"
-    else:
-        print("[bench] WARNING: no GPU visible, running on CPU (debug only)",
-              file=sys.stderr)
-        device = "cpu"
-
-    from aggregathor_amd import experiments, ops
+def bench_one(args, gar, f, device, rank, n_gpus):
+    """Build an engine for one GAR config, run warmup + timed steps, return
+    the contract result dict (or None on non-zero ranks)."""
+    from aggregathor_amd import experiments
     from aggregathor_amd.graph import Engine
     from aggregathor_amd.parallel import WorkerGroup
-
-    if device.startswith("cuda") and not ops.hip_available():
-        raise RuntimeError("HIP extension _gar_hip not built -- run "
-                           "`python -m aggregathor_amd.ops.build` first")
 
     image_size = args.image_size if args.dataset == "imagenet" else 0
     exp = experiments.instantiate(
@@ -99,8 +68,8 @@ def main():
         from aggregathor_amd.attacks.lossy import LossyChannel
         lossy = LossyChannel(args.lossy)
     engine = Engine(
-        exp, args.gar, group, nbbyzwrks=args.f, amp=amp,
-        nb_real_byz=(args.f if args.attack else 0), attack=args.attack,
+        exp, gar, group, nbbyzwrks=f, amp=amp,
+        nb_real_byz=(f if args.attack else 0), attack=args.attack,
         optimizer="sgd", learning_rate="fixed", graph_warmup=1, lossy=lossy)
 
     def sync():
@@ -134,10 +103,11 @@ def main():
 
     steps_per_sec = args.steps / elapsed
     global_batch = args.batch_size * args.workers
+    result = None
     if rank == 0:
         result = {
-            "metric": f"steps/sec {args.model} n={args.workers} f={args.f} "
-                      f"{args.gar}",
+            "metric": f"steps/sec {args.model} n={args.workers} f={f} "
+                      f"{gar}",
             "value": steps_per_sec,
             "unit": "steps/s",
             "n_gpus": n_gpus,
@@ -154,15 +124,83 @@ def main():
                 "global_batch": global_batch,
                 "seq_len": args.image_size,
                 "parallelism": f"dp{n_gpus}",
-                "gar": args.gar,
+                "gar": gar,
                 "n_workers": args.workers,
-                "f": args.f,
+                "f": f,
                 "attack": args.attack or None,
                 "images_per_sec": steps_per_sec * global_batch,
                 "final_loss": final_loss,
             },
         }
-        print(json.dumps(result), flush=True)
+    # Release this engine's HBM (graphs, activations, gradient matrix)
+    # before the next GAR config builds its own.
+    del engine, exp, lossy
+    gc.collect()
+    if device.startswith("cuda"):
+        torch.cuda.empty_cache()
+    return result
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", type=str, default="resnet50")
+    ap.add_argument("--dataset", type=str, default="imagenet",
+                    help="dataset shape: imagenet | cifar10")
+    ap.add_argument("--workers", type=int, default=8, help="total GAR workers n")
+    ap.add_argument("--f", type=int, default=2, help="declared Byzantine workers")
+    ap.add_argument("--gar", type=str, default="krum",
+                    help="aggregation rule (krum | bulyan | average | ... | "
+                         "all = the BASELINE comparison: krum f=2, bulyan "
+                         "f=1 [n >= 4f+3 bound at n=8], average f=0)")
+    ap.add_argument("--batch-size", type=int, default=32, help="per-worker batch")
+    ap.add_argument("--image-size", type=int, default=224)
+    ap.add_argument("--attack", type=str, default="",
+                    help="optional attack (e.g. reversal) mounted by f workers")
+    ap.add_argument("--lossy", nargs="*", default=None,
+                    help="UDP-style lossy-channel injection key:value args")
+    ap.add_argument("--device", type=str, default="",
+                    help="override device (debug; cpu allowed)")
+    ap.add_argument("--no-amp", action="store_true",
+                    help="disable bf16 autocast (fp32 compute)")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world)
+
+    if args.device:
+        device = args.device
+    elif torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        print("[bench] WARNING: no GPU visible, running on CPU (debug only)",
+              file=sys.stderr)
+        device = "cpu"
+
+    from aggregathor_amd import ops
+    if device.startswith("cuda") and not ops.hip_available():
+        raise RuntimeError("HIP extension _gar_hip not built -- run "
+                           "`python -m aggregathor_amd.ops.build` first")
+
+    if args.gar == "all":
+        # The BASELINE.json comparison, one driver command: same n, same
+        # per-step work, three GARs (Bulyan at f=1: n >= 4f+3 requires
+        # n >= 11 for f=2, which no GPU count divides; see module docstring).
+        configs = [("krum", args.f),
+                   ("bulyan", min(args.f, max((args.workers - 3) // 4, 0))),
+                   ("average", 0)]
+    else:
+        configs = [(args.gar, args.f)]
+
+    for gar, f in configs:
+        result = bench_one(args, gar, f, device, rank, n_gpus)
+        if result is not None:
+            print(json.dumps(result), flush=True)
 
 
 if __name__ == "__main__":

@@ -135,6 +135,13 @@ def main():
         if args.use_tpu or args.reuse_tpu:
             tools.warning("--use-tpu/--reuse-tpu: no TPU on this platform "
                           "(ignored)")
+        if args.no_wait:
+            # Reference semantics (runner.py:601-610): without --no-wait a
+            # --server process parked as a cluster node after training. Here
+            # no process outlives its training run (ranks exit together), so
+            # --no-wait is always in effect.
+            tools.warning("--no-wait: processes never park as cluster nodes "
+                          "in the RCCL model (always in effect)")
         tools.print_args("experiment", args.experiment, args.experiment_args or [])
         tools.print_args("aggregator", args.aggregator, args.aggregator_args or [])
 
