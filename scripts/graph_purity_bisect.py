@@ -25,7 +25,8 @@ os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 import torch
 import torch.nn as nn
 
-torch.backends.cudnn.benchmark = True
+torch.backends.cudnn.benchmark = \
+    os.environ.get("AGGREGATHOR_BISECT_BENCHMARK", "1") == "1"
 if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
     torch.backends.cudnn.deterministic = True
 
@@ -144,8 +145,8 @@ def model_cases():
             return getattr(m, attr)
         return make
 
-    def whole(arch):
-        return lambda: NETWORKS[arch](num_classes=10)
+    def whole(arch, classes=10):
+        return lambda: NETWORKS[arch](num_classes=classes)
 
     return [
         ("r50_layer1_8px", slice_of("resnet50", "layer1"), (B, 64, 8, 8)),
@@ -157,6 +158,28 @@ def model_cases():
     ]
 
 
+def imagenet_cases():
+    """The flagship bench shapes (resnet50-imagenet, batch 32/worker):
+    stage inputs at 56/28/14/7 px -- the shapes the headline number rides."""
+    from aggregathor_amd.models import NETWORKS
+    B = 32
+
+    def slice_of(arch, attr):
+        def make():
+            m = NETWORKS[arch](num_classes=1000)
+            return getattr(m, attr)
+        return make
+
+    return [
+        ("IM_r50_layer1_56px", slice_of("resnet50", "layer1"), (B, 64, 56, 56)),
+        ("IM_r50_layer2_28px", slice_of("resnet50", "layer2"), (B, 256, 56, 56)),
+        ("IM_r50_layer3_14px", slice_of("resnet50", "layer3"), (B, 512, 28, 28)),
+        ("IM_r50_layer4_7px", slice_of("resnet50", "layer4"), (B, 1024, 14, 14)),
+        ("IM_resnet50_full", lambda: NETWORKS["resnet50"](num_classes=1000),
+         (B, 3, 224, 224)),
+    ]
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--filter", default="")
@@ -164,14 +187,21 @@ def main():
     ap.add_argument("--views", default="1")
     ap.add_argument("--channels-last", dest="cl", default="1")
     ap.add_argument("--replays", type=int, default=3)
+    ap.add_argument("--set", dest="case_set", default="cifar",
+                    choices=["cifar", "imagenet", "all"])
     args = ap.parse_args()
     amp = args.amp == "1"
     views = args.views == "1"
     cl = args.cl == "1"
-    print(f"config: amp={amp} grad_views={views} channels_last={cl}",
-          flush=True)
+    print(f"config: amp={amp} grad_views={views} channels_last={cl} "
+          f"set={args.case_set}", flush=True)
+    cases = []
+    if args.case_set in ("cifar", "all"):
+        cases += conv_bn_cases() + model_cases()
+    if args.case_set in ("imagenet", "all"):
+        cases += imagenet_cases()
     drifted = []
-    for name, make, shape in conv_bn_cases() + model_cases():
+    for name, make, shape in cases:
         if args.filter and args.filter not in name:
             continue
         try:
