@@ -238,6 +238,11 @@ class Engine:
             self.use_graphs = CapturedStep.supported(self)
         else:
             self.use_graphs = bool(use_graphs) and CapturedStep.supported(self)
+        if self.use_graphs:
+            # No conv has executed yet (first forward is in step()): the
+            # capture-unsafe solver exclusion can still take effect.
+            from .parallel.graphstep import enable_graph_safe_conv
+            enable_graph_safe_conv()
         self._graphstep = None
 
     # ------------------------------------------------------------------ #

@@ -22,9 +22,42 @@ scalar lr), no lossy channel (host-side mask generation), and only
 graph-safe attacks (pure device math). The engine falls back to eager
 whenever the gate fails -- bitwise-identical semantics either way, which
 the GPU test asserts.
+
+CAPTURE SAFETY (round-2 root cause): MIOpen's implicit-GEMM conv solvers
+(ConvHipImplicitGemmGroup{Fwd,Wrw}Xdlops on ROCm 7.x) are NOT replay-pure
+under hipGraph capture -- replaying the captured local phase twice on
+frozen inputs drifts the produced gradients by up to ~1e33 / inf
+(bisected per-op in profiles/graph_purity_bisect.md; training on affected
+shapes diverges after ~170 steps). Two defenses, both on by default:
+  1. `enable_graph_safe_conv()` disables the offending solver family via
+     MIOpen env vars BEFORE the first convolution runs (call it early --
+     bench.py/runner.py do so whenever graphs may engage);
+  2. `_record()` runs a replay-purity SELF-CHECK after capture: the local
+     graph is replayed twice on identical inputs and the gradient rows
+     compared -- non-finite or large drift discards the graphs and falls
+     back to eager execution (correctness over speed), small nonzero
+     drift (atomic accumulation order) is allowed with a warning.
 """
 
+import os
+
 import torch
+
+from .. import tools
+
+
+def enable_graph_safe_conv():
+    """Exclude capture-unsafe MIOpen conv solvers (idempotent; must run
+    before the first conv executes in the process).
+
+    Evidence: with MIOPEN_DEBUG_CONV_IMPLICIT_GEMM=0 every per-op and
+    full-model replay-purity case is bitwise pure and 400 captured
+    resnet50-cifar10 steps train healthy (vs divergence at step ~172 with
+    the family enabled) -- see profiles/graph_purity_bisect.md.
+    """
+    if os.environ.get("AGGREGATHOR_UNSAFE_SOLVERS") == "1":
+        return  # measurement escape hatch: keep MIOpen defaults
+    os.environ.setdefault("MIOPEN_DEBUG_CONV_IMPLICIT_GEMM", "0")
 
 
 def _attack_graph_safe(attack):
@@ -105,7 +138,8 @@ class CapturedStep:
         return self.run()
 
     def _record(self):
-        """Record both graphs (recording executes nothing)."""
+        """Record both graphs (recording executes nothing), then verify
+        the recording is replay-pure before trusting it."""
         eng = self.engine
         self.graph_local = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph_local):
@@ -113,7 +147,48 @@ class CapturedStep:
         self.graph_apply = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph_apply):
             eng._apply_from_matrix()
+        self._verify_replay_purity()
         self.ready = True
+
+    def _verify_replay_purity(self):
+        """Replay the captured local phase twice on identical inputs: the
+        produced gradient rows must match. A capture-unsafe kernel (state
+        prepared outside the captured sequence, e.g. a lazily-zeroed
+        solver workspace) shows up here as non-finite values or large
+        drift -- raise so the engine falls back to eager execution. Small
+        nonzero drift is nondeterministic-but-correct accumulation order;
+        allowed (it cannot break cross-rank bit-identity, which depends
+        only on the GATHERED matrix being identical everywhere)."""
+        if os.environ.get("AGGREGATHOR_NO_PURITY_CHECK") == "1":
+            return
+        eng = self.engine
+        self._stage_batches()
+        self.graph_local.replay()
+        torch.cuda.synchronize()
+        ref = eng.local_rows.clone()
+        self.graph_local.replay()
+        torch.cuda.synchronize()
+        cur = eng.local_rows
+        finite = bool(torch.isfinite(cur).all()) and \
+            bool(torch.isfinite(ref).all())
+        if not finite:
+            raise RuntimeError(
+                "captured step is not replay-pure (non-finite gradients on "
+                "re-replay): a capture-unsafe kernel is in the graph")
+        drift = (cur - ref).abs().max().item()
+        del ref
+        if drift == 0.0:
+            return
+        scale = max(cur.abs().max().item(), 1e-12)
+        if drift > 1e-3 * scale:
+            raise RuntimeError(
+                f"captured step is not replay-pure (drift {drift:.3e} vs "
+                f"scale {scale:.3e}): a capture-unsafe kernel is in the graph")
+        tools.warning(
+            f"captured step has bounded nondeterministic drift "
+            f"({drift:.3e}, scale {scale:.3e}): atomic accumulation order; "
+            f"runs are not bitwise reproducible (cross-rank identity is "
+            f"unaffected)")
 
     def _local_phase(self):
         """The capturable worker loop, reading the static batch buffers."""

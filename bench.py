@@ -40,12 +40,18 @@ import time
 os.environ.pop("MIOPEN_FIND_MODE", None)
 os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 
+from aggregathor_amd.parallel.graphstep import enable_graph_safe_conv
+
+# The bench rides hipGraph capture: exclude the capture-unsafe MIOpen
+# implicit-GEMM solvers (round-2 root cause -- graphstep.py docstring,
+# profiles/graph_purity_bisect.md) before the first conv runs.
+enable_graph_safe_conv()
+
 import torch
 
 torch.backends.cudnn.benchmark = True
 if os.environ.get("AGGREGATHOR_DETERMINISTIC_CONV") == "1":
-    # Excludes atomic-accumulation conv algorithms (candidate fix for the
-    # hipGraph open issue, NOTES.md).
+    # Excludes atomic-accumulation conv algorithms (kept as a debug switch).
     torch.backends.cudnn.deterministic = True
 
 

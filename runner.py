@@ -103,11 +103,13 @@ parser.add_argument("--trace", action="store_true", default=False,
                          "step execution")
 parser.add_argument("--amp", action="store_true", default=False,
                     help="bf16 autocast compute (fp32 gradients/aggregation)")
-parser.add_argument("--graphs", action="store_true", default=False,
-                    help="hipGraph step capture (opt-in here: long runs with "
-                         "the evaluation service showed divergence on some "
-                         "configs, under investigation -- see NOTES.md; "
-                         "bench.py uses graphs on its validated configs)")
+parser.add_argument("--graphs", type=str, default="auto",
+                    choices=["auto", "on", "off"],
+                    help="hipGraph step capture (default auto: on for "
+                         "capturable configs, with capture-unsafe MIOpen "
+                         "solvers excluded and a replay-purity self-check "
+                         "at capture that falls back to eager on failure "
+                         "-- the round-2 root-cause fix, NOTES.md)")
 parser.add_argument("--seed", type=int, default=1234)
 parser.add_argument("--profile-steps", type=int, default=0,
                     help="Profile this many steps with torch.profiler and "
@@ -144,6 +146,12 @@ def main():
                           "in the RCCL model (always in effect)")
         tools.print_args("experiment", args.experiment, args.experiment_args or [])
         tools.print_args("aggregator", args.aggregator, args.aggregator_args or [])
+
+    if args.graphs in ("auto", "on"):
+        # Must precede the first convolution: excludes capture-unsafe
+        # MIOpen solvers (graphstep.py docstring).
+        from aggregathor_amd.parallel.graphstep import enable_graph_safe_conv
+        enable_graph_safe_conv()
 
     import torch
     from aggregathor_amd import experiments
@@ -194,7 +202,7 @@ def main():
             nb_real_byz=args.nb_real_byz_workers, attack=args.attack,
             attack_args=args.attack_args or [], lossy=lossy, amp=args.amp,
             trace=args.trace, seed=args.seed, integrity=integrity,
-            use_graphs=("auto" if args.graphs else False))
+            use_graphs=("auto" if args.graphs == "auto" else args.graphs == "on"))
         tools.info(f"model d = {engine.d} parameters, GAR = {args.aggregator}")
 
     with tools.Context("session", "info"):

@@ -9,6 +9,13 @@ non-finite step plus the max |param| trajectory, to localize the corruption
 
 Usage (on a GPU box):
   python scripts/graph_divergence_repro.py [--steps 400] [--case all]
+
+ROOT CAUSE FOUND (round 2): MIOpen's ConvHipImplicitGemmGroup*Xdlops
+solvers are not replay-pure under hipGraph capture (see
+scripts/graph_purity_bisect.py and profiles/graph_purity_bisect.md).
+Engine now auto-excludes them when graphs engage, so REPRODUCING the
+historical divergence requires AGGREGATHOR_UNSAFE_SOLVERS=1 (and
+AGGREGATHOR_NO_PURITY_CHECK=1 to get past the capture self-check).
 """
 
 import argparse
