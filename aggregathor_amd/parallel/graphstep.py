@@ -28,15 +28,15 @@ CAPTURE SAFETY (round-2 root cause): MIOpen's implicit-GEMM conv solvers
 under hipGraph capture -- replaying the captured local phase twice on
 frozen inputs drifts the produced gradients by up to ~1e33 / inf
 (bisected per-op in profiles/graph_purity_bisect.md; training on affected
-shapes diverges after ~170 steps). Two defenses, both on by default:
-  1. `enable_graph_safe_conv()` disables the offending solver family via
-     MIOpen env vars BEFORE the first convolution runs (call it early --
-     bench.py/runner.py do so whenever graphs may engage);
-  2. `_record()` runs a replay-purity SELF-CHECK after capture: the local
-     graph is replayed twice on identical inputs and the gradient rows
-     compared -- non-finite or large drift discards the graphs and falls
-     back to eager execution (correctness over speed), small nonzero
-     drift (atomic accumulation order) is allowed with a warning.
+shapes diverges after ~170 steps). Defense on by default:
+`_record()` runs a replay-purity SELF-CHECK after capture -- the local
+graph is replayed twice on identical inputs and the gradient rows
+compared; non-finite or large drift discards the graphs and falls back
+to eager execution (correctness over speed), small nonzero drift (atomic
+accumulation order) is allowed with a warning. Optionally,
+AGGREGATHOR_SAFE_SOLVERS=1 (enable_graph_safe_conv) excludes the solver
+family process-wide -- guaranteed-pure captures, measured as the right
+choice only for small-spatial configs (see the function docstring).
 """
 
 import os
@@ -47,17 +47,26 @@ from .. import tools
 
 
 def enable_graph_safe_conv():
-    """Exclude capture-unsafe MIOpen conv solvers (idempotent; must run
-    before the first conv executes in the process).
+    """Optionally exclude capture-unsafe MIOpen conv solvers (idempotent;
+    must run before the first conv executes in the process).
 
-    Evidence: with MIOPEN_DEBUG_CONV_IMPLICIT_GEMM=0 every per-op and
-    full-model replay-purity case is bitwise pure and 400 captured
-    resnet50-cifar10 steps train healthy (vs divergence at step ~172 with
-    the family enabled) -- see profiles/graph_purity_bisect.md.
+    Measured trade-off (gpurun_out/bench_*.log, MI355X):
+      * With MIOPEN_DEBUG_CONV_IMPLICIT_GEMM=0 every per-op and full-model
+        replay-purity case is bitwise pure and 400 captured
+        resnet50-cifar10 steps train healthy (divergence at step ~172
+        otherwise) -- but on the IMAGENET-shape flagship MIOpen then falls
+        back to naive conv: 13,015 ms/step vs 68 (unusable).
+      * A WRW-only exclusion is insufficient (still impure).
+    Therefore the DEFAULT leaves MIOpen's solver choice untouched and
+    relies on the capture-time replay-purity self-check: a pure capture
+    keeps the hipGraph fast path (68 ms/step), an impure one falls back
+    to eager (78 ms/step) -- both correct. Set AGGREGATHOR_SAFE_SOLVERS=1
+    to force the exclusion process-wide (guaranteed-pure captures; only
+    sensible for small-spatial configs where the non-igemm solvers are
+    competitive).
     """
-    if os.environ.get("AGGREGATHOR_UNSAFE_SOLVERS") == "1":
-        return  # measurement escape hatch: keep MIOpen defaults
-    os.environ.setdefault("MIOPEN_DEBUG_CONV_IMPLICIT_GEMM", "0")
+    if os.environ.get("AGGREGATHOR_SAFE_SOLVERS") == "1":
+        os.environ.setdefault("MIOPEN_DEBUG_CONV_IMPLICIT_GEMM", "0")
 
 
 def _attack_graph_safe(attack):

@@ -42,9 +42,10 @@ os.environ.setdefault("PYTORCH_MIOPEN_SUGGEST_NHWC", "1")
 
 from aggregathor_amd.parallel.graphstep import enable_graph_safe_conv
 
-# The bench rides hipGraph capture: exclude the capture-unsafe MIOpen
-# implicit-GEMM solvers (round-2 root cause -- graphstep.py docstring,
-# profiles/graph_purity_bisect.md) before the first conv runs.
+# hipGraph capture safety: the capture-time replay-purity self-check
+# (graphstep.py docstring, profiles/graph_purity_bisect.md) gates the
+# graph fast path; AGGREGATHOR_SAFE_SOLVERS=1 additionally excludes the
+# capture-unsafe MIOpen solver family (small-shape configs only).
 enable_graph_safe_conv()
 
 import torch

@@ -148,8 +148,9 @@ def main():
         tools.print_args("aggregator", args.aggregator, args.aggregator_args or [])
 
     if args.graphs in ("auto", "on"):
-        # Must precede the first convolution: excludes capture-unsafe
-        # MIOpen solvers (graphstep.py docstring).
+        # Must precede the first convolution: honors the opt-in
+        # AGGREGATHOR_SAFE_SOLVERS exclusion (graphstep.py docstring); the
+        # capture-time replay-purity self-check is the default gate.
         from aggregathor_amd.parallel.graphstep import enable_graph_safe_conv
         enable_graph_safe_conv()
 

@@ -13,9 +13,9 @@ Usage (on a GPU box):
 ROOT CAUSE FOUND (round 2): MIOpen's ConvHipImplicitGemmGroup*Xdlops
 solvers are not replay-pure under hipGraph capture (see
 scripts/graph_purity_bisect.py and profiles/graph_purity_bisect.md).
-Engine now auto-excludes them when graphs engage, so REPRODUCING the
-historical divergence requires AGGREGATHOR_UNSAFE_SOLVERS=1 (and
-AGGREGATHOR_NO_PURITY_CHECK=1 to get past the capture self-check).
+The engine now runs a replay-purity self-check at capture and falls back
+to eager on failure, so REPRODUCING the historical divergence requires
+AGGREGATHOR_NO_PURITY_CHECK=1.
 """
 
 import argparse
