@@ -160,24 +160,48 @@ class CapturedStep:
         self.ready = True
 
     def _verify_replay_purity(self):
-        """Replay the captured local phase twice on identical inputs: the
-        produced gradient rows must match. A capture-unsafe kernel (state
-        prepared outside the captured sequence, e.g. a lazily-zeroed
-        solver workspace) shows up here as non-finite values or large
-        drift -- raise so the engine falls back to eager execution. Small
-        nonzero drift is nondeterministic-but-correct accumulation order;
-        allowed (it cannot break cross-rank bit-identity, which depends
-        only on the GATHERED matrix being identical everywhere)."""
+        """Run the captured step CYCLE twice on identical state and compare
+        the produced gradient rows. A capture-unsafe kernel (state prepared
+        outside the captured sequence, e.g. a lazily-zeroed solver
+        workspace) shows up as non-finite values or large drift -- raise so
+        the engine falls back to eager execution. Small nonzero drift is
+        nondeterministic-but-correct accumulation order; allowed (it cannot
+        break cross-rank bit-identity, which depends only on the GATHERED
+        matrix being identical everywhere).
+
+        The check reproduces the REAL per-step sequence (local graph ->
+        gather -> apply graph -> stage -> local graph), because kernel
+        state dependencies can differ between back-to-back replays and the
+        interleaved production cycle. Model params/buffers and optimizer
+        state are snapshotted and rewound, so the check is trajectory-
+        transparent: training continues exactly as if it never ran."""
         if os.environ.get("AGGREGATHOR_NO_PURITY_CHECK") == "1":
             return
         eng = self.engine
-        self._stage_batches()
-        self.graph_local.replay()
-        torch.cuda.synchronize()
-        ref = eng.local_rows.clone()
-        self.graph_local.replay()
-        torch.cuda.synchronize()
-        cur = eng.local_rows
+        params = [p.detach() for p in eng.params]
+        buffers = list(eng.model.buffers())
+        opt_tensors = [t for s in eng.optimizer.state.values()
+                       for t in s.values() if torch.is_tensor(t)]
+        snap = [t.clone() for t in params + buffers + opt_tensors]
+        try:
+            self._stage_batches()
+            self.graph_local.replay()
+            torch.cuda.synchronize()
+            ref = eng.local_rows.clone()
+            # The interleaving work of a real step.
+            if eng.group.distributed:
+                eng.group.gather(eng.local_rows, out=eng.matrix)
+            self.graph_apply.replay()
+            # Rewind the model so the second cycle sees identical inputs.
+            for t, s in zip(params + buffers, snap):
+                t.copy_(s)
+            self._stage_batches()
+            self.graph_local.replay()
+            torch.cuda.synchronize()
+            cur = eng.local_rows.clone()
+        finally:
+            for t, s in zip(params + buffers + opt_tensors, snap):
+                t.copy_(s)
         finite = bool(torch.isfinite(cur).all()) and \
             bool(torch.isfinite(ref).all())
         if not finite:
@@ -185,7 +209,6 @@ class CapturedStep:
                 "captured step is not replay-pure (non-finite gradients on "
                 "re-replay): a capture-unsafe kernel is in the graph")
         drift = (cur - ref).abs().max().item()
-        del ref
         if drift == 0.0:
             return
         scale = max(cur.abs().max().item(), 1e-12)
