@@ -82,6 +82,11 @@ def _slim_alias(arch):
 
 for _arch in NETWORKS:
     for _ds in _DATASETS:
+        if min(_DATASETS[_ds]["shape"][1:]) < getattr(
+                NETWORKS[_arch], "min_input", 0):
+            continue  # e.g. inception_v3 needs >= 75 px (models/__init__)
         cls = _make(_arch, _ds)
         register(f"{_arch}-{_ds}", cls)
-        register(f"slim-{_slim_alias(_arch)}-{_ds}", cls)
+        alias = f"slim-{_slim_alias(_arch)}-{_ds}"
+        if alias != f"{_arch}-{_ds}":
+            register(alias, cls)

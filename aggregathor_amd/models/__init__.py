@@ -18,6 +18,12 @@ from .resnet import (resnet18, resnet34, resnet50, resnet101, resnet152,
                      cifar_resnet56, cifar_resnet110, RESNETS)
 from .vgg import vgg11, vgg16, vgg19
 from .mobilenet import mobilenet_v2
+from .inception import inception_v1, inception_v3
+
+# Inception v3's padding-free grid reductions need a real spatial extent
+# (the canonical input is 299; >= 75 keeps every grid >= 1). The
+# experiment registry skips dataset shapes below this.
+inception_v3.min_input = 75
 
 #: Full model-factory map (the reference's slim ``networks_map`` analog,
 #: external/slim/nets/nets_factory.py:39-66).
@@ -25,6 +31,7 @@ NETWORKS = dict(RESNETS)
 NETWORKS.update({
     "vgg11": vgg11, "vgg16": vgg16, "vgg19": vgg19,
     "mobilenet_v2": mobilenet_v2,
+    "inception_v1": inception_v1, "inception_v3": inception_v3,
 })
 
 __all__ = [
@@ -32,4 +39,5 @@ __all__ = [
     "resnet18", "resnet34", "resnet50", "resnet101", "resnet152", "resnet200",
     "cifar_resnet20", "cifar_resnet32", "cifar_resnet44", "cifar_resnet56",
     "cifar_resnet110", "vgg11", "vgg16", "vgg19", "mobilenet_v2",
+    "inception_v1", "inception_v3",
 ]

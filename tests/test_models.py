@@ -77,3 +77,36 @@ def test_extra_families_through_engine(exp_name):
     for _ in range(2):
         loss = eng.step()
     assert loss == loss  # finite
+
+
+def test_inception_v1_forward_and_engine():
+    import torch
+    from aggregathor_amd.models import inception_v1
+    m = inception_v1(num_classes=10)
+    y = m(torch.randn(2, 3, 32, 32))
+    assert y.shape == (2, 10)
+    # Engine path on the registered experiment.
+    from aggregathor_amd import experiments
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    exp = experiments.instantiate(
+        "inception_v1-cifar10", ["batch-size:2", "eval-examples:0"])
+    eng = Engine(exp, "median", WorkerGroup(3), nbbyzwrks=1)
+    assert eng.step() == eng.last_loss
+
+
+def test_inception_v3_forward_and_registration():
+    import torch
+    from aggregathor_amd import experiments
+    from aggregathor_amd.models import inception_v3
+    m = inception_v3(num_classes=7)
+    m.eval()
+    with torch.no_grad():
+        y = m(torch.randn(1, 3, 96, 96))
+    assert y.shape == (1, 7)
+    names = experiments.itemize()
+    assert "inception_v3-imagenet" in names
+    assert "slim-inception_v3-imagenet" in names
+    # v3's padding-free reductions cannot run on 32 px: not registered.
+    assert "inception_v3-cifar10" not in names
+    assert "inception_v1-cifar10" in names
