@@ -19,7 +19,8 @@ class CNNetExperiment(_Experiment):
     def __init__(self, args):
         args = tools.parse_keyval(args, defaults={
             "batch-size": 32, "eval-batch-size": 1024, "seed": 1234,
-            "eval-examples": 1024, "data-dir": "", "data-pool": 8})
+            "eval-examples": 1024, "data-dir": "", "data-pool": 8,
+            "signal": 0.5})
         if args["batch-size"] <= 0:
             raise tools.UserException("Cannot make batches of non-positive size")
         self.args = args
@@ -30,7 +31,7 @@ class CNNetExperiment(_Experiment):
         self._synth = SyntheticClassification(
             (3, 32, 32), 10, seed=args["seed"],
             eval_examples=args["eval-examples"],
-            pool_size=args["data-pool"])
+            pool_size=args["data-pool"], signal=args["signal"])
 
     def model(self):
         return CNNet()

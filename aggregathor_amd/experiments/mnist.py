@@ -40,7 +40,8 @@ class MNIST(_Experiment):
     def __init__(self, args):
         args = tools.parse_keyval(args, defaults={
             "batch-size": 32, "eval-batch-size": 1024, "seed": 1234,
-            "data-dir": "", "eval-examples": 1024, "data-pool": 8})
+            "data-dir": "", "eval-examples": 1024, "data-pool": 8,
+            "signal": 0.5})
         if args["batch-size"] <= 0:
             raise tools.UserException("Cannot make batches of non-positive size")
         self.args = args
@@ -59,7 +60,7 @@ class MNIST(_Experiment):
                 raise tools.UserException(f"Cannot load MNIST from {data_dir!r}: {e}")
         self._synth = SyntheticClassification(
             (784,), 10, seed=args["seed"], eval_examples=args["eval-examples"],
-            pool_size=args["data-pool"])
+            pool_size=args["data-pool"], signal=args["signal"])
 
     def model(self):
         return MLP((784, 100, 10))

@@ -26,7 +26,7 @@ class ResNetExperiment(_Experiment):
         args = tools.parse_keyval(args, defaults={
             "batch-size": 32, "eval-batch-size": 256, "seed": 1234,
             "eval-examples": 512, "image-size": 0, "data-dir": "",
-            "data-pool": 8})
+            "data-pool": 8, "signal": 0.5})
         if args["batch-size"] <= 0:
             raise tools.UserException("Cannot make batches of non-positive size")
         self.args = args
@@ -49,7 +49,7 @@ class ResNetExperiment(_Experiment):
         self._synth = SyntheticClassification(
             shape, self.classes, seed=args["seed"],
             eval_examples=args["eval-examples"],
-            pool_size=args["data-pool"])
+            pool_size=args["data-pool"], signal=args["signal"])
 
     def train_batch(self, worker, step, device):
         src = self._real if self._real is not None else self._synth
