@@ -68,9 +68,16 @@ def build_graph(make_module, input_shape, amp=True):
 
 
 def run_case(name, make_module, input_shape):
+    # Preallocate the comparison buffers BEFORE capture: an allocation
+    # anywhere after capture is itself a candidate trigger.
+    import contextlib
+    from aggregathor_amd.graph import flat_size
+    probe_model = make_module()
+    d = flat_size([p for p in probe_model.parameters() if p.requires_grad])
+    del probe_model
+    ref1 = torch.empty(d, device="cuda:0")
+    ref2 = torch.empty(d, device="cuda:0")
     g, row = build_graph(make_module, input_shape)
-    ref1 = torch.empty_like(row)
-    ref2 = torch.empty_like(row)
 
     def drift(protocol):
         g.replay()
