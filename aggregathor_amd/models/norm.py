@@ -5,6 +5,17 @@ measured ~30% of the ResNet-50 training step (profiles/). PyTorch's native
 batch-norm path (cudnn/miopen disabled for the op) uses fewer, fused
 kernels. ``AGGREGATHOR_BN=native|miopen`` selects the implementation; the
 default is the measured winner on MI355X.
+
+STATUS of the from-scratch kernels (AGGREGATHOR_BN=fused,
+ops/csrc/bn_kernels.hip): NUMERICS TESTBED, not a hot-path component.
+After three measured optimization rounds (NOTES.md) they reached MIOpen
+PARITY (0.112 vs 0.103 ms/layer-pass; 69.1 vs 67.5 ms full step) but not
+a win, so the default stays ``miopen`` and the fused path is kept for
+what it uniquely provides: a fully-controlled, deterministic,
+fp64-referenced BatchNorm implementation used by the numerics tests
+(tests/test_gpu_bn.py) -- including the shifted-variance formulation that
+documents (and guards against) the catastrophic E[x^2]-E[x]^2
+cancellation. It is not selected by any default configuration.
 """
 
 import os
