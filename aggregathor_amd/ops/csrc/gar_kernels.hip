@@ -306,6 +306,112 @@ __global__ __launch_bounds__(kBlock) void sqdist_tile_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// LDS-staged single-HBM-pass kernel for 16 < n <= 64 (round-2 fix for the
+// large-n bandwidth cliff, profiles/gar_scaling_n.txt).
+//
+// The 8x8 tile kernel above re-reads each row chunk ~n/8 times from HBM
+// because its pair tiles land on different XCDs whose L2s cannot share the
+// working set (measured: krum algorithmic bandwidth 4.4 TB/s at n=8 but
+// 1.75 at n=16 and 0.93 at n=32). This kernel makes the amplification 1:
+// each workgroup stages a [n, CHUNK] tile of the matrix in LDS ONCE per
+// chunk, then its 4 waves sweep all upper-triangular 8x8 pair tiles from
+// LDS, so every gradient byte is fetched from HBM exactly once per block
+// stripe. Per chunk, each wave reduces its 64 per-lane pair partials with
+// a fixed-shape shfl_xor butterfly (deterministic) and accumulates them
+// into an LDS pair table; the table is flushed to the per-block partials
+// row at the end (same layout as the other kernels, reduced by
+// sqdist_reduce_kernel). No atomics anywhere; every reduction order is
+// fixed -> bit-deterministic, replicated-GAR safe.
+//
+// LDS budget (64 KB default): tile n*CHUNK*4 B + pair table
+// n(n-1)/2*4 B. CHUNK is chosen at launch as the largest multiple of 64
+// that fits (n=32 -> 448 cols, n=64 -> 192).
+
+template <int VW>
+__global__ __launch_bounds__(kBlock) void sqdist_lds_kernel(
+    const float* __restrict__ g, float* __restrict__ partials, long d,
+    int n, int chunk, int tiles_per_row, int ntiles) {
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  const int P = n * (n - 1) / 2;
+  float* tile = lds;          // [n, chunk] staged rows
+  float* accp = lds + (long)n * chunk;  // [P] pair accumulators
+
+  for (int p = threadIdx.x; p < P; p += kBlock) accp[p] = 0.f;
+
+  const int lane = (int)threadIdx.x & 63;
+  const int wave = (int)threadIdx.x >> 6;
+  const int nwaves = kBlock / 64;
+  const long nchunks = (d + chunk - 1) / chunk;
+
+  for (long ci = blockIdx.x; ci < nchunks; ci += gridDim.x) {
+    const long c0 = ci * chunk;
+    const int cw = (int)((d - c0) < chunk ? (d - c0) : chunk);
+    __syncthreads();  // pair-table reads of the previous chunk done
+    // Stage [n, cw] into LDS (vectorized when alignment allows: chunk is
+    // a multiple of 64, so c0 keeps the row alignment class of d).
+    const int cwv = cw / VW;
+    for (int idx = threadIdx.x; idx < n * cwv; idx += kBlock) {
+      int r = idx / cwv, cc = (idx - r * cwv) * VW;
+      fvec<VW> v = *reinterpret_cast<const fvec<VW>*>(g + (long)r * d + c0 + cc);
+#pragma unroll
+      for (int k = 0; k < VW; ++k) tile[(long)r * chunk + cc + k] = v.v[k];
+    }
+    for (int idx = threadIdx.x; idx < n * (cw - cwv * VW); idx += kBlock) {
+      int rem = cw - cwv * VW;
+      int r = idx / rem, cc = cwv * VW + (idx - r * rem);
+      tile[(long)r * chunk + cc] = g[(long)r * d + c0 + cc];
+    }
+    __syncthreads();
+
+    for (int t = wave; t < ntiles; t += nwaves) {
+      // Decode upper-triangular tile t -> (ti, tj), ti <= tj.
+      int ti = 0, rem = t;
+      while (rem >= tiles_per_row - ti) {
+        rem -= tiles_per_row - ti;
+        ++ti;
+      }
+      int tj = ti + rem;
+      int i0 = ti * 8, j0 = tj * 8;
+      float acc[64];
+#pragma unroll
+      for (int p = 0; p < 64; ++p) acc[p] = 0.f;
+      for (int x = lane; x < cw; x += 64) {
+        float vi[8], vj[8];
+#pragma unroll
+        for (int a = 0; a < 8; ++a) {
+          vi[a] = (i0 + a < n) ? tile[(long)(i0 + a) * chunk + x] : 0.f;
+          vj[a] = (j0 + a < n) ? tile[(long)(j0 + a) * chunk + x] : 0.f;
+        }
+#pragma unroll
+        for (int a = 0; a < 8; ++a)
+#pragma unroll
+          for (int b = 0; b < 8; ++b) {
+            float dd = vi[a] - vj[b];
+            acc[a * 8 + b] = fmaf(dd, dd, acc[a * 8 + b]);
+          }
+      }
+      // Fixed-shape butterfly: every lane ends with the wave total.
+#pragma unroll
+      for (int p = 0; p < 64; ++p)
+#pragma unroll
+        for (int s = 32; s > 0; s >>= 1)
+          acc[p] += __shfl_xor(acc[p], s, 64);
+      // One lane per pair folds into the LDS table (tile owned by this
+      // wave alone -> no race; chunk-sequential order -> deterministic).
+#pragma unroll
+      for (int p = 0; p < 64; ++p) {
+        int gi = i0 + (p >> 3), gj = j0 + (p & 7);
+        if (lane == (p & 63) && gi < n && gj < n && gi < gj)
+          accp[pair_index(gi, gj, n)] += acc[p];
+      }
+    }
+  }
+  __syncthreads();
+  for (int p = threadIdx.x; p < P; p += kBlock)
+    partials[(long)blockIdx.x * P + p] = accp[p];
+}
+
+// ---------------------------------------------------------------------------
 // MFMA pairwise-distance kernel (n <= 16): the matrix-core formulation of
 // the same computation.
 //
@@ -495,16 +601,38 @@ void sqdist(const float* g, int n, long d, float* partials, float* dist,
       sqdist_small_kernel<16, 1>
           <<<nblk, kBlock, 0, stream>>>(g, partials, d, d, n);
   } else {
-    // measured: the 8x8 tile at VW=4 (170 VGPR + scratch) loses to VW=2.
     int tiles_per_row = (n + 7) / 8;
     int ntiles = tiles_per_row * (tiles_per_row + 1) / 2;
-    dim3 grid(nblk, ntiles);
-    if (vw >= 2)
-      sqdist_tile_kernel<2><<<grid, kBlock, 0, stream>>>(g, partials, d / 2,
-                                                         d, n, tiles_per_row);
-    else
-      sqdist_tile_kernel<1><<<grid, kBlock, 0, stream>>>(g, partials, d, d,
-                                                         n, tiles_per_row);
+    const char* e = getenv("AGGREGATHOR_SQDIST");
+    if (e && strcmp(e, "tile") == 0) {
+      // Legacy multi-pass tile kernel (A/B reference; ~n/8 HBM re-reads).
+      // Measured: VW=4 (170 VGPR + scratch) loses to VW=2.
+      dim3 grid(nblk, ntiles);
+      if (vw >= 2)
+        sqdist_tile_kernel<2><<<grid, kBlock, 0, stream>>>(
+            g, partials, d / 2, d, n, tiles_per_row);
+      else
+        sqdist_tile_kernel<1><<<grid, kBlock, 0, stream>>>(g, partials, d, d,
+                                                           n, tiles_per_row);
+    } else {
+      // Default: LDS-staged single-HBM-pass kernel.
+      int P = n * (n - 1) / 2;
+      int chunk = ((64 * 1024 / 4 - P) / n) / 64 * 64;
+      long nchunks = (d + chunk - 1) / chunk;
+      int grid = (int)(nchunks < 512 ? nchunks : 512);
+      if (grid > nblk) grid = nblk;
+      nblk = grid;
+      size_t shmem = ((size_t)n * chunk + P) * 4;
+      if (vw >= 4)
+        sqdist_lds_kernel<4><<<grid, kBlock, shmem, stream>>>(
+            g, partials, d, n, chunk, tiles_per_row, ntiles);
+      else if (vw >= 2)
+        sqdist_lds_kernel<2><<<grid, kBlock, shmem, stream>>>(
+            g, partials, d, n, chunk, tiles_per_row, ntiles);
+      else
+        sqdist_lds_kernel<1><<<grid, kBlock, shmem, stream>>>(
+            g, partials, d, n, chunk, tiles_per_row, ntiles);
+    }
   }
   int P = n * (n - 1) / 2;
   sqdist_reduce_kernel<<<P, kBlock, 0, stream>>>(partials, dist, nblk, n,

@@ -32,7 +32,8 @@ def _ref64(fn, g_cpu, *args):
 
 
 @pytest.mark.parametrize("n,d", [(4, 1000), (8, 100003), (8, 1 << 20),
-                                 (16, 65537), (24, 10000), (33, 4099)])
+                                 (16, 65537), (24, 10000), (33, 4099),
+                                 (32, 1 << 20), (64, (1 << 20) + 1)])
 def test_sqdist(n, d):
     ext = _require_ext()
     g = _rand(n, d, seed=n)
@@ -216,3 +217,22 @@ def test_gpu_identical_rows():
     torch.testing.assert_close(out, torch.ones(100000), rtol=1e-6, atol=0)
     out = ext.bulyan(torch.ones((11, 50000), device="cuda"), 2, 7).cpu()
     torch.testing.assert_close(out, torch.ones(50000), rtol=1e-6, atol=0)
+
+
+@pytest.mark.parametrize("n", [24, 32, 64])
+def test_sqdist_lds_kernel_deterministic_and_matches_tile(n):
+    # The round-2 LDS-staged single-pass kernel (n > 16) must be bitwise
+    # deterministic run-to-run; and its sums must agree with the legacy
+    # multi-pass tile kernel to fp32 reduction tolerance.
+    import os
+    import subprocess
+    import sys
+    ext = _require_ext()
+    d = 1 << 18
+    g = _rand(n, d, seed=n + 100)
+    a = ext.pairwise_sqdist(g).cpu()
+    b = ext.pairwise_sqdist(g).cpu()
+    assert torch.equal(a, b), "LDS sqdist kernel is not deterministic"
+    want = _ref64(R.pairwise_sqdist, g.cpu())
+    off = ~torch.eye(n, dtype=torch.bool)
+    torch.testing.assert_close(a[off], want[off], rtol=1e-5, atol=1e-3)
