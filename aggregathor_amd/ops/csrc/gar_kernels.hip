@@ -327,8 +327,10 @@ __global__ __launch_bounds__(kBlock) void sqdist_tile_kernel(
 // n(n-1)/2*4 B. CHUNK is chosen at launch as the largest multiple of 64
 // that fits (n=32 -> 448 cols, n=64 -> 192).
 
+constexpr int kLdsBlock = 512;  // 8 waves: latency hiding at 1 block/CU
+
 template <int VW>
-__global__ __launch_bounds__(kBlock) void sqdist_lds_kernel(
+__global__ __launch_bounds__(kLdsBlock) void sqdist_lds_kernel(
     const float* __restrict__ g, float* __restrict__ partials, long d,
     int n, int chunk, int tiles_per_row, int ntiles) {
   extern __shared__ __attribute__((aligned(16))) float lds[];
@@ -336,11 +338,11 @@ __global__ __launch_bounds__(kBlock) void sqdist_lds_kernel(
   float* tile = lds;          // [n, chunk] staged rows
   float* accp = lds + (long)n * chunk;  // [P] pair accumulators
 
-  for (int p = threadIdx.x; p < P; p += kBlock) accp[p] = 0.f;
+  for (int p = threadIdx.x; p < P; p += kLdsBlock) accp[p] = 0.f;
 
   const int lane = (int)threadIdx.x & 63;
   const int wave = (int)threadIdx.x >> 6;
-  const int nwaves = kBlock / 64;
+  const int nwaves = kLdsBlock / 64;
   const long nchunks = (d + chunk - 1) / chunk;
 
   for (long ci = blockIdx.x; ci < nchunks; ci += gridDim.x) {
@@ -350,13 +352,13 @@ __global__ __launch_bounds__(kBlock) void sqdist_lds_kernel(
     // Stage [n, cw] into LDS (vectorized when alignment allows: chunk is
     // a multiple of 64, so c0 keeps the row alignment class of d).
     const int cwv = cw / VW;
-    for (int idx = threadIdx.x; idx < n * cwv; idx += kBlock) {
+    for (int idx = threadIdx.x; idx < n * cwv; idx += kLdsBlock) {
       int r = idx / cwv, cc = (idx - r * cwv) * VW;
       fvec<VW> v = *reinterpret_cast<const fvec<VW>*>(g + (long)r * d + c0 + cc);
 #pragma unroll
       for (int k = 0; k < VW; ++k) tile[(long)r * chunk + cc + k] = v.v[k];
     }
-    for (int idx = threadIdx.x; idx < n * (cw - cwv * VW); idx += kBlock) {
+    for (int idx = threadIdx.x; idx < n * (cw - cwv * VW); idx += kLdsBlock) {
       int rem = cw - cwv * VW;
       int r = idx / rem, cc = cwv * VW + (idx - r * rem);
       tile[(long)r * chunk + cc] = g[(long)r * d + c0 + cc];
@@ -390,24 +392,33 @@ __global__ __launch_bounds__(kBlock) void sqdist_lds_kernel(
             acc[a * 8 + b] = fmaf(dd, dd, acc[a * 8 + b]);
           }
       }
-      // Fixed-shape butterfly: every lane ends with the wave total.
+      // Butterfly TRANSPOSE-reduce (fixed shape, deterministic): at the
+      // step with xor-distance s the register count halves and pair-index
+      // bit log2(s) is re-encoded as lane bit log2(s); after 6 steps
+      // lane l's acc[0] holds the complete wave sum of pair index l.
 #pragma unroll
-      for (int p = 0; p < 64; ++p)
+      for (int s = 32, R = 32; s >= 1; s >>= 1, R >>= 1) {
 #pragma unroll
-        for (int s = 32; s > 0; s >>= 1)
-          acc[p] += __shfl_xor(acc[p], s, 64);
-      // One lane per pair folds into the LDS table (tile owned by this
-      // wave alone -> no race; chunk-sequential order -> deterministic).
-#pragma unroll
-      for (int p = 0; p < 64; ++p) {
-        int gi = i0 + (p >> 3), gj = j0 + (p & 7);
-        if (lane == (p & 63) && gi < n && gj < n && gi < gj)
-          accp[pair_index(gi, gj, n)] += acc[p];
+        for (int p = 0; p < 32; ++p) {
+          if (p >= R) break;
+          bool hi = (lane & s) != 0;
+          float mine = hi ? acc[p + R] : acc[p];
+          float send = hi ? acc[p] : acc[p + R];
+          acc[p] = mine + __shfl_xor(send, s, 64);
+        }
+      }
+      // One parallel LDS add per lane (distinct pairs -> distinct
+      // addresses; tile owned by this wave alone -> no race; chunk-
+      // sequential order -> deterministic).
+      {
+        int gi = i0 + (lane >> 3), gj = j0 + (lane & 7);
+        if (gi < n && gj < n && gi < gj)
+          accp[pair_index(gi, gj, n)] += acc[0];
       }
     }
   }
   __syncthreads();
-  for (int p = threadIdx.x; p < P; p += kBlock)
+  for (int p = threadIdx.x; p < P; p += kLdsBlock)
     partials[(long)blockIdx.x * P + p] = accp[p];
 }
 
@@ -615,22 +626,31 @@ void sqdist(const float* g, int n, long d, float* partials, float* dist,
         sqdist_tile_kernel<1><<<grid, kBlock, 0, stream>>>(g, partials, d, d,
                                                            n, tiles_per_row);
     } else {
-      // Default: LDS-staged single-HBM-pass kernel.
+      // Default: LDS-staged single-HBM-pass kernel. 128 KB dynamic LDS
+      // (gfx950 has 160 KB/CU; >64 KB needs the explicit opt-in) doubles
+      // the chunk, halving the per-chunk fold overhead; occupancy is one
+      // block/CU, which the 8-wave block size covers.
       int P = n * (n - 1) / 2;
-      int chunk = ((64 * 1024 / 4 - P) / n) / 64 * 64;
+      int chunk = ((128 * 1024 / 4 - P) / n) / 64 * 64;
       long nchunks = (d + chunk - 1) / chunk;
       int grid = (int)(nchunks < 512 ? nchunks : 512);
       if (grid > nblk) grid = nblk;
       nblk = grid;
       size_t shmem = ((size_t)n * chunk + P) * 4;
+      const void* fn = (vw >= 4) ? (const void*)&sqdist_lds_kernel<4>
+                     : (vw >= 2) ? (const void*)&sqdist_lds_kernel<2>
+                                 : (const void*)&sqdist_lds_kernel<1>;
+      (void)hipFuncSetAttribute(fn,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)shmem);
       if (vw >= 4)
-        sqdist_lds_kernel<4><<<grid, kBlock, shmem, stream>>>(
+        sqdist_lds_kernel<4><<<grid, kLdsBlock, shmem, stream>>>(
             g, partials, d, n, chunk, tiles_per_row, ntiles);
       else if (vw >= 2)
-        sqdist_lds_kernel<2><<<grid, kBlock, shmem, stream>>>(
+        sqdist_lds_kernel<2><<<grid, kLdsBlock, shmem, stream>>>(
             g, partials, d, n, chunk, tiles_per_row, ntiles);
       else
-        sqdist_lds_kernel<1><<<grid, kBlock, shmem, stream>>>(
+        sqdist_lds_kernel<1><<<grid, kLdsBlock, shmem, stream>>>(
             g, partials, d, n, chunk, tiles_per_row, ntiles);
     }
   }
