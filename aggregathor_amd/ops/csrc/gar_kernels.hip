@@ -377,7 +377,40 @@ __global__ __launch_bounds__(kLdsBlock) void sqdist_lds_kernel(
       float acc[64];
 #pragma unroll
       for (int p = 0; p < 64; ++p) acc[p] = 0.f;
-      for (int x = lane; x < cw; x += 64) {
+      // Main loop: ds_read_b128 (4 coords per LDS access per row); chunk
+      // is a multiple of 64 so LDS rows are 16 B aligned.
+      const int cw4 = cw & ~3;
+      for (int x = lane * 4; x < cw4; x += 64 * 4) {
+        fvec<4> vi[8], vj[8];
+#pragma unroll
+        for (int a = 0; a < 8; ++a) {
+          if (i0 + a < n)
+            vi[a] = *reinterpret_cast<const fvec<4>*>(
+                tile + (long)(i0 + a) * chunk + x);
+          else
+#pragma unroll
+            for (int k = 0; k < 4; ++k) vi[a].v[k] = 0.f;
+          if (j0 + a < n)
+            vj[a] = *reinterpret_cast<const fvec<4>*>(
+                tile + (long)(j0 + a) * chunk + x);
+          else
+#pragma unroll
+            for (int k = 0; k < 4; ++k) vj[a].v[k] = 0.f;
+        }
+#pragma unroll
+        for (int a = 0; a < 8; ++a)
+#pragma unroll
+          for (int b = 0; b < 8; ++b) {
+            float s = acc[a * 8 + b];
+#pragma unroll
+            for (int k = 0; k < 4; ++k) {
+              float dd = vi[a].v[k] - vj[b].v[k];
+              s = fmaf(dd, dd, s);
+            }
+            acc[a * 8 + b] = s;
+          }
+      }
+      for (int x = cw4 + lane; x < cw; x += 64) {  // scalar tail
         float vi[8], vj[8];
 #pragma unroll
         for (int a = 0; a < 8; ++a) {
