@@ -60,3 +60,22 @@ def test_bench_under_torchrun_world2():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["n_workers"] == 8  # strong scaling: n fixed
+
+
+def test_bench_gar_all_emits_three_lines():
+    # BASELINE.json's "Krum & Bulyan vs average" in one driver command:
+    # one contract JSON line per GAR, same n, documented bulyan f bound.
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--model", "resnet20", "--dataset",
+         "cifar10", "--steps", "2", "--warmup", "1", "--batch-size", "2",
+         "--device", "cpu", "--gar", "all"],
+        capture_output=True, timeout=900, env=_clean_env(), cwd=str(REPO))
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    lines = [l for l in r.stdout.decode().splitlines() if l.startswith("{")]
+    assert len(lines) == 3
+    rows = [json.loads(l) for l in lines]
+    assert [x["config"]["gar"] for x in rows] == ["krum", "bulyan", "average"]
+    assert [x["config"]["f"] for x in rows] == [2, 1, 0]
+    for x in rows:
+        assert REQUIRED_KEYS <= set(x)
+        assert x["config"]["n_workers"] == 8

@@ -84,3 +84,27 @@ def test_bulyan_engine_gpu():
     for _ in range(5):
         loss = eng.step()
     assert math.isfinite(loss)
+
+
+@pytest.mark.gpu
+def test_capture_self_check_engages_hybrid_on_unsafe_shapes():
+    # resnet50-cifar10 is the historically-diverging config: the capture
+    # self-check must detect the capture-unsafe conv solvers and engage
+    # HYBRID mode (eager local phase + bitwise-verified apply graph) --
+    # or, if this ROCm stack's solvers happen to verify pure, keep full
+    # graphs. Either way training must proceed finite; what is FORBIDDEN
+    # is an unverified full-capture (the round-1 corruption).
+    from aggregathor_amd import experiments
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    exp = experiments.instantiate(
+        "resnet50-cifar10", ["batch-size:4", "eval-examples:0"])
+    eng = Engine(exp, "average", WorkerGroup(8, device="cuda:0"),
+                 amp=True, use_graphs=True, graph_warmup=1)
+    losses = [eng.step() for _ in range(8)]
+    assert all(l == l for l in losses), losses
+    if eng.use_graphs:
+        gs = eng._graphstep
+        assert gs is not None and gs.ready
+        assert gs.mode in ("full", "apply")
+    # else: self-check rejected even the apply graph -> full eager (valid).
