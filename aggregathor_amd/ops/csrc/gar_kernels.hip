@@ -1070,11 +1070,30 @@ __global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
     const float* __restrict__ g, float* __restrict__ out, long dv, long d,
     int n, int t, int b, const unsigned char* __restrict__ flags_g,
     const float* __restrict__ inv_mk) {
-  __shared__ unsigned char fl[NMAX * NMAX];
   __shared__ float inv[NMAX];
-  for (int i = threadIdx.x; i < t * n; i += blockDim.x) fl[i] = flags_g[i];
-  if ((int)threadIdx.x < t) inv[threadIdx.x] = inv_mk[threadIdx.x];
+  // Selection sets as uint64 bitmasks (n <= 64) in LDS, read back through
+  // a VOLATILE pointer inside the hot loop: the per-wave SGPR file (~102
+  // words) cannot hold t masks as hoisted uniforms -- a byte-matrix
+  // version produced 633 SGPR spills + 272 B/lane scratch (measured 33x
+  // over the kernel's traffic bound), and a register-array version 2522
+  // spills. The volatile ds_read_b64 lands each mask in VGPRs per
+  // iteration (t reads per coordinate, trivially cheap) and the bit
+  // tests stay VALU cndmask/add chains with zero scratch.
+  __shared__ unsigned long long mk_s[NMAX];
+  if ((int)threadIdx.x < t) {
+    inv[threadIdx.x] = inv_mk[threadIdx.x];
+    unsigned long long m = 0;
+    for (int i = 0; i < n; ++i)
+      if (flags_g[threadIdx.x * n + i]) m |= 1ull << i;
+    mk_s[threadIdx.x] = m;
+  }
   __syncthreads();
+  // The masks must stay in VGPRs, re-read per iteration: left to itself
+  // the compiler hoists the (wave-uniform) loads into SGPRs and spills
+  // them (~102 addressable SGPRs/wave; the byte-matrix version measured
+  // 633 spills + 272 B/lane scratch, 33x over the traffic bound). The
+  // empty volatile asm pins each half in a VGPR and is opaque to LICM.
+  const unsigned* mkv32 = reinterpret_cast<const unsigned*>(mk_s);
 
   const long stride = (long)gridDim.x * blockDim.x;
   for (long x = (long)blockIdx.x * blockDim.x + threadIdx.x; x < dv;
@@ -1092,11 +1111,14 @@ __global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
 #pragma unroll
       for (int k = 0; k < NMAX; ++k) {
         if (k >= t) continue;
+        unsigned mlo = mkv32[2 * k], mhi = mkv32[2 * k + 1];
+        asm volatile("" : "+v"(mlo), "+v"(mhi));
+        const unsigned long long m = ((unsigned long long)mhi << 32) | mlo;
         float s = 0.f;
 #pragma unroll
         for (int i = 0; i < NMAX; ++i) {
           if (i >= n) continue;
-          if (fl[k * n + i]) s += colv[i].v[c];
+          if ((m >> i) & 1) s += colv[i].v[c];
         }
         inters[k] = s * inv[k];
       }
@@ -1111,11 +1133,14 @@ __global__ __launch_bounds__(kBlock) void bulyan_final_kernel(
 #pragma unroll
     for (int k = 0; k < NMAX; ++k) {
       if (k >= t) continue;
+      unsigned mlo = mkv32[2 * k], mhi = mkv32[2 * k + 1];
+      asm volatile("" : "+v"(mlo), "+v"(mhi));
+      const unsigned long long m = ((unsigned long long)mhi << 32) | mlo;
       float s = 0.f;
 #pragma unroll
       for (int i = 0; i < NMAX; ++i) {
         if (i >= n) continue;
-        if (fl[k * n + i]) s += g[(long)i * d + x];
+        if ((m >> i) & 1) s += g[(long)i * d + x];
       }
       inters[k] = s * inv[k];
     }
