@@ -203,6 +203,7 @@ class Engine:
 
         self.phase_times = {}
         self._eval_model = None
+        self._eval_stream = None
         # Serializes training steps against the service threads' device work
         # (evaluate / state_dict): concurrent eval-model builds + MIOpen
         # benchmark-find + forwards interleaving with hipGraph replays were
@@ -439,14 +440,41 @@ class Engine:
         rank 0 only -- a broadcast here would deadlock the other ranks).
         BN running stats are the local rank's, exactly like the reference's
         eval replicas reading the PS variables concurrently with updates.
-        """
-        with self.lock:
-            if self.device.type == "cuda":
-                torch.cuda.synchronize()
-            if self._eval_model is None:
-                self._eval_model = self.experiment.model().to(self.device)
+
+        Default: serialized with training via the engine lock (pauses
+        training for the eval duration -- the sound choice on a shared
+        GPU). AGGREGATHOR_CONCURRENT_EVAL=1 restores the reference's
+        concurrent-eval capability (runner.py:318-330 allocated separate
+        eval DEVICES) on a dedicated HIP stream WITHOUT the lock -- the
+        eval replica may then read mid-step (torn) parameters, exactly
+        like the reference's eval replicas reading PS variables during
+        updates. Gated to runs with no live captured graphs: eval-side
+        device allocations permanently corrupt captured MIOpen kernels
+        (the round-1 eval-corruption finding, root-caused in
+        profiles/graph_purity_bisect.md)."""
+        import os
+        concurrent = (os.environ.get("AGGREGATHOR_CONCURRENT_EVAL") == "1"
+                      and not self.use_graphs and self._graphstep is None)
+        if not concurrent:
+            with self.lock:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                if self._eval_model is None:
+                    self._eval_model = self.experiment.model().to(self.device)
+                self._eval_model.load_state_dict(self.model.state_dict())
+                return self.experiment.accuracy(self._eval_model, self.device)
+        if self._eval_model is None:
+            self._eval_model = self.experiment.model().to(self.device)
+        if self.device.type != "cuda":
             self._eval_model.load_state_dict(self.model.state_dict())
             return self.experiment.accuracy(self._eval_model, self.device)
+        if self._eval_stream is None:
+            self._eval_stream = torch.cuda.Stream()
+        with torch.cuda.stream(self._eval_stream):
+            self._eval_model.load_state_dict(self.model.state_dict())
+            metrics = self.experiment.accuracy(self._eval_model, self.device)
+        self._eval_stream.synchronize()
+        return metrics
 
     def state_dict(self):
         with self.lock:

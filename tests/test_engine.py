@@ -231,3 +231,37 @@ def test_cnnet_learns():
         loss = eng.step()
     assert math.isfinite(loss)
     assert eng.evaluate()["top1-X-acc"] > 0.3
+
+
+def test_concurrent_eval_env(monkeypatch):
+    # AGGREGATHOR_CONCURRENT_EVAL=1: evaluate() skips the engine lock when
+    # no captured graphs are live (reference concurrent-eval capability,
+    # runner.py:318-330). On CPU this exercises the gating logic and the
+    # torn-read-tolerant path; the GPU test covers the stream variant.
+    import threading
+    from aggregathor_amd import experiments
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    monkeypatch.setenv("AGGREGATHOR_CONCURRENT_EVAL", "1")
+    exp = experiments.instantiate("mnist", ["batch-size:8",
+                                            "eval-examples:64"])
+    eng = Engine(exp, "average", WorkerGroup(2))
+    # evaluate() must work while the lock is HELD by a trainer thread
+    # (i.e. it really does not take the lock).
+    with eng.lock:
+        metrics = eng.evaluate()
+    assert 0.0 <= metrics["top1-X-acc"] <= 1.0
+    # Interleaved with steps.
+    errs = []
+    def stepper():
+        try:
+            for _ in range(5):
+                eng.step()
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+    t = threading.Thread(target=stepper)
+    t.start()
+    for _ in range(3):
+        eng.evaluate()
+    t.join()
+    assert not errs

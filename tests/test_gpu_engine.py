@@ -108,3 +108,31 @@ def test_capture_self_check_engages_hybrid_on_unsafe_shapes():
         assert gs is not None and gs.ready
         assert gs.mode in ("full", "apply")
     # else: self-check rejected even the apply graph -> full eager (valid).
+
+
+@pytest.mark.gpu
+def test_concurrent_eval_gpu(monkeypatch):
+    # Concurrent eval on a dedicated stream, no engine lock, eager mode.
+    import threading
+    from aggregathor_amd import experiments
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    monkeypatch.setenv("AGGREGATHOR_CONCURRENT_EVAL", "1")
+    exp = experiments.instantiate(
+        "resnet20-cifar10", ["batch-size:8", "eval-examples:64"])
+    eng = Engine(exp, "krum", WorkerGroup(4, device="cuda:0"),
+                 nbbyzwrks=1, use_graphs=False)
+    errs = []
+    def stepper():
+        try:
+            for _ in range(10):
+                eng.step()
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+    t = threading.Thread(target=stepper)
+    t.start()
+    accs = [eng.evaluate()["top1-X-acc"] for _ in range(3)]
+    t.join()
+    assert not errs
+    assert all(0.0 <= a <= 1.0 for a in accs)
+    assert eng.last_loss == eng.last_loss  # training stayed finite
