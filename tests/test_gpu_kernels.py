@@ -236,3 +236,34 @@ def test_sqdist_lds_kernel_deterministic_and_matches_tile(n):
     want = _ref64(R.pairwise_sqdist, g.cpu())
     off = ~torch.eye(n, dtype=torch.bool)
     torch.testing.assert_close(a[off], want[off], rtol=1e-5, atol=1e-3)
+
+
+@pytest.mark.parametrize("n", [24, 33, 64])
+def test_sqdist_nan_propagates_large_n(n):
+    # NaN handling through the LDS-staged large-n kernel: a NaN coordinate
+    # in row r must make every distance involving r NaN (ordered last by
+    # the selection's total order), exactly like the small-n kernels.
+    ext = _require_ext()
+    g = _rand(n, 100000, seed=n)
+    g[n // 2, 31337] = float("nan")
+    got = ext.pairwise_sqdist(g).cpu()
+    for j in range(n):
+        if j != n // 2:
+            assert torch.isnan(got[n // 2, j]) and torch.isnan(got[j, n // 2])
+    # Healthy pairs stay finite.
+    assert torch.isfinite(got[0, 1])
+
+
+def test_gpu_large_n_bulyan_with_nan_row():
+    # Bulyan at n > 16 with a NaN-poisoned row: the distance pass orders it
+    # last, selection excludes it, and the bitmask-folded final pass must
+    # produce a finite aggregate matching the fp64 oracle.
+    ext = _require_ext()
+    n, f = 24, 3
+    m = n - f - 2
+    g = _rand(n, 30000, seed=7)
+    g[2] = float("nan")
+    got = ext.bulyan(g, f, m).cpu()
+    assert torch.isfinite(got).all()
+    want = _ref64(R.bulyan, g.cpu(), f, m)
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-6)
