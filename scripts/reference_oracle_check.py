@@ -220,7 +220,8 @@ def make_cases():
     """(name, g, f) input matrix: randoms, NaN-laced, outliers, ties."""
     rng = np.random.default_rng(7)
     cases = []
-    for n, d in ((5, 7), (8, 100), (8, 1000), (11, 257), (16, 64)):
+    for n, d in ((5, 7), (8, 100), (8, 1000), (11, 257), (16, 64),
+                 (24, 301), (32, 129), (64, 67)):
         cases.append((f"normal_n{n}_d{d}",
                       rng.standard_normal((n, d)).astype(np.float32), 2))
     g = rng.standard_normal((8, 64)).astype(np.float32)
