@@ -46,6 +46,7 @@ class WorkerGroup:
         self.local_workers = nbworkers // self.world
         self.worker_ids = list(range(self.rank * self.local_workers,
                                      (self.rank + 1) * self.local_workers))
+        self._flag_buf = None  # lazy; see any_rank
 
     @property
     def distributed(self):
@@ -113,10 +114,16 @@ class WorkerGroup:
         next ``gather`` until the process-group timeout. Every rank folds
         its local flag in every step so all ranks agree on the abort step
         (local losses differ per rank, so local detection steps would too).
+
+        The flag buffer is PREALLOCATED: this runs every step, and steady-
+        state device allocations are the trigger class that corrupts live
+        captured graphs (profiles/graph_purity_bisect.md).
         """
         if self.world == 1:
             return bool(flag)
-        dev = self.device if self.backend == "nccl" else "cpu"
-        t = torch.tensor([1.0 if flag else 0.0], device=dev)
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        return bool(t.item())
+        if self._flag_buf is None:
+            dev = self.device if self.backend == "nccl" else "cpu"
+            self._flag_buf = torch.zeros(1, device=dev)
+        self._flag_buf.fill_(1.0 if flag else 0.0)
+        dist.all_reduce(self._flag_buf, op=dist.ReduceOp.MAX)
+        return bool(self._flag_buf.item())
