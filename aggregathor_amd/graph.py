@@ -246,6 +246,20 @@ class Engine:
             enable_graph_safe_conv()
         self._graphstep = None
 
+        # Opt-in bucketed gather overlap (parallel/overlap.py): hide the
+        # all-gather behind backward. Scope v1: one worker per rank,
+        # eager local phase, no real-Byzantine attack (it rewrites the
+        # row after backward, when buckets would already be in flight).
+        self.overlap = None
+        import os as _os
+        _mb = _os.environ.get("AGGREGATHOR_BUCKET_MB")
+        if (_mb and group.distributed and group.local_workers == 1
+                and not self.use_graphs and self.attack is None):
+            from .parallel.overlap import BucketedGather
+            self.overlap = BucketedGather(
+                group, self.params, self.local_rows, self.matrix,
+                bucket_bytes=int(float(_mb) * (1 << 20)))
+
     # ------------------------------------------------------------------ #
 
     def _trace(self, msg):
@@ -289,6 +303,8 @@ class Engine:
         # One autocast region for the whole worker loop: the bf16 weight
         # casts are cached across the local workers' forwards instead of
         # being re-materialized per micro-batch.
+        if self.overlap is not None:
+            self.overlap.begin_step()
         amp_ctx = (torch.autocast(device_type="cuda", dtype=torch.bfloat16)
                    if self.amp else contextlib.nullcontext())
         with amp_ctx:
@@ -333,12 +349,18 @@ class Engine:
             row.copy_(self.attack.craft(row.clone(), worker,
                                         self.global_step))
 
+    def _gather_matrix(self):
+        if self.overlap is not None:
+            return self.overlap.finish()
+        return self.group.gather(
+            self.local_rows, out=self.matrix if self.group.distributed else None)
+
     def aggregate(self, matrix=None):
         """Gather all rows (unless pre-gathered), verify integrity, inject
         channel loss, run the GAR."""
         if matrix is None:
             self._trace("gather")
-            matrix = self.group.gather(self.local_rows, out=self.matrix if self.group.distributed else None)
+            matrix = self._gather_matrix()
         if self.integrity is not None:
             self._trace("integrity check")
             macs_local = self.integrity.sign_rows(
@@ -412,9 +434,7 @@ class Engine:
             t = time.monotonic()
             loss = self.compute_local_gradients()
             t = self._phase("local_gradients", t)
-            matrix = self.group.gather(
-                self.local_rows,
-                out=self.matrix if self.group.distributed else None)
+            matrix = self._gather_matrix()
             t = self._phase("gather", t)
             aggregated = self.aggregate(matrix)
             t = self._phase("aggregate", t)

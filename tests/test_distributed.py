@@ -118,3 +118,81 @@ def test_world2_collective_nan_abort(tmp_path):
     assert metas[0]["steps"] == metas[1]["steps"] == 4
     # Clean exit well under the process-group timeout.
     assert wall < 60, f"abort took {wall:.1f}s -- ranks likely hung in gather"
+
+
+def test_world2_bucketed_gather_bit_identical(tmp_path):
+    # Opt-in bucketed gather overlap (parallel/overlap.py): world-2, one
+    # worker per rank, small buckets (many collectives per step). The
+    # final parameters must be BITWISE identical to the serial-gather
+    # world-2 run and to the single-process run.
+    env_extra = {"AGGREGATHOR_BUCKET_MB": "0.05"}
+    single = _run_single(tmp_path, steps=6, aggregator="average", n=2, f=0)
+    r0, r1 = _run_world(tmp_path, 2, steps=6, aggregator="average", n=2,
+                        f=0, port=29690)
+    # Re-run world-2 WITH bucketed gather.
+    procs, outs = [], []
+    for rank in range(2):
+        out = tmp_path / f"bucket_rank{rank}.pt"
+        outs.append(out)
+        env = dict(os.environ)
+        env.update({
+            "RANK": str(rank), "WORLD_SIZE": "2", "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29691",
+            "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME", "lo"),
+        })
+        env.update(env_extra)
+        procs.append(subprocess.Popen(
+            [sys.executable, str(WORKER), str(out), "6", "average", "2",
+             "0", ""],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=300)
+        assert p.returncode == 0, f"worker failed:\n{stderr.decode()[-2000:]}"
+    b0 = torch.load(outs[0], weights_only=False)
+    b1 = torch.load(outs[1], weights_only=False)
+    assert torch.equal(b0["flat"], b1["flat"])
+    assert torch.equal(b0["flat"], r0["flat"])
+    assert torch.equal(b0["flat"], single["flat"])
+
+
+def test_bucketed_gather_unit_world1():
+    # Bucket layout + flush-at-finish semantics at world 1 (gloo identity
+    # gather): finish() without any hook having fired must still gather
+    # every bucket, and the scattered matrix must equal the row bitwise.
+    import torch.distributed as dist
+    import torch.nn as nn
+    from aggregathor_amd.graph import bind_grad_views, flat_size
+    from aggregathor_amd.parallel.overlap import BucketedGather
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29692")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        class _G:
+            distributed = True
+            local_workers = 1
+        model = nn.Sequential(nn.Linear(17, 33), nn.ReLU(),
+                              nn.Linear(33, 5))
+        params = list(model.parameters())
+        d = flat_size(params)
+        row = torch.zeros((1, d))
+        matrix = torch.empty((1, d))
+        bg = BucketedGather(_G(), params, row, matrix, bucket_bytes=64)
+        assert len(bg.spans) >= 3  # tiny buckets -> several spans
+        assert bg.spans[0][1] == d and bg.spans[-1][0] == 0  # full cover
+        # Path 1: flush-only (no hooks fired).
+        bg.begin_step()
+        row.uniform_(-1, 1)
+        out = bg.finish()
+        assert torch.equal(out[0], row[0])
+        # Path 2: hooks fire through a real backward.
+        bg.begin_step()
+        row.zero_()
+        bind_grad_views(params, row[0])
+        loss = model(torch.randn(4, 17)).square().mean()
+        loss.backward()
+        out = bg.finish()
+        assert torch.equal(out[0], row[0])
+        assert out.abs().sum() > 0
+        bg.remove()
+    finally:
+        dist.destroy_process_group()
