@@ -28,11 +28,12 @@ CAPTURE SAFETY (round-2 root cause): MIOpen's implicit-GEMM conv solvers
 under hipGraph capture -- replaying the captured local phase twice on
 frozen inputs drifts the produced gradients by up to ~1e33 / inf
 (bisected per-op in profiles/graph_purity_bisect.md; training on affected
-shapes diverges after ~170 steps). Defense on by default:
-`_record()` runs a replay-purity SELF-CHECK after capture -- the local
-graph is replayed twice on identical inputs and the gradient rows
-compared; non-finite or large drift discards the graphs and falls back
-to eager execution (correctness over speed), small nonzero drift (atomic
+shapes diverges after ~170 steps). Defense on by default: `_record()`
+runs a SELF-CHECK after capture -- graph-vs-eager correctness and
+replay-to-replay stability in relative L2, trajectory-transparent via
+full state rewind. On failure the engine first tries HYBRID mode (eager
+local phase + the apply graph, verified BITWISE replay-pure -- it has no
+conv kernels), and only then full eager. Small nonzero drift (atomic
 accumulation order) is allowed with a warning. Optionally,
 AGGREGATHOR_SAFE_SOLVERS=1 (enable_graph_safe_conv) excludes the solver
 family process-wide -- guaranteed-pure captures, measured as the right
