@@ -196,3 +196,17 @@ def test_bucketed_gather_unit_world1():
         bg.remove()
     finally:
         dist.destroy_process_group()
+
+
+def test_bucketed_gather_gating(monkeypatch):
+    # The overlap must NOT engage outside its v1 scope: single-process
+    # runs, multiple local workers, or real-Byzantine attacks.
+    from aggregathor_amd import experiments
+    from aggregathor_amd.graph import Engine
+    from aggregathor_amd.parallel import WorkerGroup
+    monkeypatch.setenv("AGGREGATHOR_BUCKET_MB", "1")
+    exp = experiments.instantiate("mnist", ["batch-size:8"])
+    # Single process: no distributed group -> no overlap.
+    eng = Engine(exp, "average", WorkerGroup(4))
+    assert eng.overlap is None
+    assert eng.step() == eng.last_loss  # and training still works
